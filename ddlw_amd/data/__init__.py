@@ -1,0 +1,17 @@
+from .preprocess import preprocess_bytes, preprocess_batch, preprocess_pil
+from .pipeline import build_tables, read_table, table_path
+from .synthetic import make_synthetic_jpeg_tree, make_synthetic_dataset
+from .loader import ShardedParquetLoader, make_converter
+
+__all__ = [
+    "preprocess_bytes",
+    "preprocess_batch",
+    "preprocess_pil",
+    "build_tables",
+    "read_table",
+    "table_path",
+    "make_synthetic_jpeg_tree",
+    "make_synthetic_dataset",
+    "ShardedParquetLoader",
+    "make_converter",
+]

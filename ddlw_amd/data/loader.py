@@ -1,0 +1,285 @@
+"""Sharded streaming Parquet loader — the Petastorm-converter equivalent.
+
+Reference contract (SURVEY.md §2.5 Petastorm row;
+``Part 1 .../03_model_training_distributed.py:135-144, 197-234, 332-337``):
+
+- ``make_converter(df_or_table)`` materialises the table into a Parquet cache
+  dir and returns a converter; ``len(converter)`` = row count (:143-144);
+- ``converter.make_torch_dataset(batch_size, cur_shard, shard_count)`` is a
+  context manager yielding an *infinite* batch stream (``num_epochs=None``
+  semantics, :199) sharded per rank;
+- ``converter.delete()`` removes the cache (:425-426).
+
+MI355X-native design (not a Petastorm port):
+
+- sharding is by Parquet **row group**: rank r owns row groups
+  r, r+W, r+2W, ... of the dataset (disjoint + exhaustive across ranks),
+  cycled infinitely — tested in ``tests/test_loader.py``;
+- a CPU worker pool decodes/praeprocesses JPEG rows into batch tensors;
+- on GPU, batches are staged through **pinned host buffers** and copied with
+  ``hipMemcpyAsync`` on a **side HIP stream** (``torch.cuda.Stream``), double
+  buffered so H2D overlaps compute; the consumer stream waits on a recorded
+  event, never the host.
+"""
+from __future__ import annotations
+
+import queue
+import threading
+import uuid
+from concurrent.futures import ThreadPoolExecutor
+from pathlib import Path
+from typing import Callable, Iterator, List, Optional, Tuple
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import torch
+
+from .preprocess import preprocess_bytes
+from ..core.config import current_setup
+
+
+def shard_row_groups(num_row_groups: int, cur_shard: int, shard_count: int) -> List[int]:
+    """Row groups owned by shard ``cur_shard`` of ``shard_count``:
+    disjoint across shards, exhaustive, round-robin (rank r gets r, r+W, ...)."""
+    if not (0 <= cur_shard < shard_count):
+        raise ValueError(f"cur_shard {cur_shard} out of range for {shard_count}")
+    return list(range(cur_shard, num_row_groups, shard_count))
+
+
+class _H2DStager:
+    """Pinned-buffer + side-stream H2D pipeline (double buffered)."""
+
+    def __init__(self, device: torch.device, depth: int = 2):
+        self.device = device
+        self.stream = torch.cuda.Stream(device=device)
+        self.depth = depth
+        self._pinned: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self._slot = 0
+
+    def stage(self, images: torch.Tensor, labels: torch.Tensor):
+        """Copy a CPU batch to the device on the side stream; returns device
+        tensors + an event the consumer stream must wait on."""
+        if not self._pinned or self._pinned[0][0].shape != images.shape:
+            self._pinned = [
+                (
+                    torch.empty_like(images, pin_memory=True),
+                    torch.empty_like(labels, pin_memory=True),
+                )
+                for _ in range(self.depth)
+            ]
+            self._events = [torch.cuda.Event() for _ in range(self.depth)]
+            self._slot = 0
+        slot = self._slot
+        self._slot = (self._slot + 1) % self.depth
+        pi, pl = self._pinned[slot]
+        ev = self._events[slot]
+        # don't overwrite a pinned buffer still being read by an in-flight copy
+        ev.synchronize()
+        pi.copy_(images)
+        pl.copy_(labels)
+        with torch.cuda.stream(self.stream):
+            di = pi.to(self.device, non_blocking=True)
+            dl = pl.to(self.device, non_blocking=True)
+            ev.record(self.stream)
+        return di, dl, ev
+
+
+class ShardedParquetLoader:
+    """Iterates (images, labels) batches from a Parquet dataset shard."""
+
+    def __init__(
+        self,
+        dataset_path: str,
+        batch_size: int = 32,
+        cur_shard: int = 0,
+        shard_count: int = 1,
+        img_height: int = 224,
+        img_width: int = 224,
+        num_epochs: Optional[int] = None,  # None = infinite (Petastorm default)
+        workers: int = 4,
+        device: Optional[torch.device] = None,
+        content_column: str = "content",
+        label_column: str = "label_idx",
+        transform: Optional[Callable] = None,
+        prefetch: int = 2,
+    ):
+        self.path = str(dataset_path)
+        self.files = sorted(str(p) for p in Path(self.path).glob("*.parquet")) or [self.path]
+        self.batch_size = batch_size
+        self.cur_shard = cur_shard
+        self.shard_count = shard_count
+        self.img_height = img_height
+        self.img_width = img_width
+        self.num_epochs = num_epochs
+        self.workers = workers
+        self.device = device
+        self.content_column = content_column
+        self.label_column = label_column
+        self.transform = transform or (
+            lambda c: preprocess_bytes(c, self.img_height, self.img_width)
+        )
+        self.prefetch = prefetch
+        # (file_idx, row_group_idx) pairs across all files
+        self._rg_index: List[Tuple[int, int]] = []
+        self._num_rows = 0
+        for fi, f in enumerate(self.files):
+            md = pq.ParquetFile(f).metadata
+            self._num_rows += md.num_rows
+            for g in range(md.num_row_groups):
+                self._rg_index.append((fi, g))
+        self._stager = None
+
+    def __len__(self) -> int:
+        return self._num_rows
+
+    # ------------------------------------------------------------------ #
+    def _shard_groups(self) -> List[Tuple[int, int]]:
+        idx = shard_row_groups(len(self._rg_index), self.cur_shard, self.shard_count)
+        return [self._rg_index[i] for i in idx]
+
+    def _iter_rows(self) -> Iterator[Tuple[bytes, int]]:
+        groups = self._shard_groups()
+        if not groups:
+            return
+        epoch = 0
+        readers = {fi: pq.ParquetFile(f) for fi, f in enumerate(self.files)}
+        while self.num_epochs is None or epoch < self.num_epochs:
+            for fi, g in groups:
+                tbl = readers[fi].read_row_group(g, columns=[self.content_column, self.label_column])
+                contents = tbl.column(self.content_column).to_pylist()
+                labels = tbl.column(self.label_column).to_pylist()
+                for c, l in zip(contents, labels):
+                    yield c, l
+            epoch += 1
+
+    def _batches_cpu(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        pool = ThreadPoolExecutor(max_workers=self.workers)
+        try:
+            rows = self._iter_rows()
+            while True:
+                chunk = []
+                for _ in range(self.batch_size):
+                    try:
+                        chunk.append(next(rows))
+                    except StopIteration:
+                        break
+                if not chunk:
+                    return
+                futs = [pool.submit(self.transform, c) for c, _ in chunk]
+                images = torch.stack([f.result() for f in futs])
+                labels = torch.tensor([l for _, l in chunk], dtype=torch.long)
+                yield images, labels
+                if len(chunk) < self.batch_size:
+                    return
+        finally:
+            pool.shutdown(wait=False)
+
+    def __iter__(self):
+        if self.device is None or self.device.type == "cpu":
+            yield from self._batches_cpu()
+            return
+        # GPU path: background decode thread + pinned double-buffer staging
+        if self._stager is None:
+            self._stager = _H2DStager(self.device, depth=max(2, self.prefetch))
+        q: "queue.Queue" = queue.Queue(maxsize=self.prefetch)
+        stop = threading.Event()
+
+        def producer():
+            try:
+                for batch in self._batches_cpu():
+                    if stop.is_set():
+                        return
+                    q.put(batch)
+            finally:
+                q.put(None)
+
+        t = threading.Thread(target=producer, daemon=True)
+        t.start()
+        try:
+            while True:
+                item = q.get()
+                if item is None:
+                    return
+                di, dl, ev = self._stager.stage(*item)
+                torch.cuda.current_stream(self.device).wait_event(ev)
+                yield di, dl
+        finally:
+            stop.set()
+            # drain so the producer can exit
+            while not q.empty():
+                try:
+                    q.get_nowait()
+                except queue.Empty:
+                    break
+
+
+class Converter:
+    """Petastorm ``SparkDatasetConverter`` equivalent over a cached Parquet
+    dataset (``make_spark_converter`` contract)."""
+
+    def __init__(self, cache_path: Path, owned: bool):
+        self.path = Path(cache_path)
+        self._owned = owned
+        self._len = sum(
+            pq.ParquetFile(f).metadata.num_rows for f in sorted(self.path.glob("*.parquet"))
+        )
+
+    def __len__(self) -> int:
+        return self._len
+
+    class _DatasetCtx:
+        def __init__(self, loader: "ShardedParquetLoader"):
+            self.loader = loader
+
+        def __enter__(self) -> "ShardedParquetLoader":
+            return self.loader
+
+        def __exit__(self, *exc) -> None:
+            pass
+
+    def make_torch_dataset(
+        self,
+        batch_size: int = 32,
+        cur_shard: int = 0,
+        shard_count: int = 1,
+        num_epochs: Optional[int] = None,
+        workers_count: int = 4,
+        device: Optional[torch.device] = None,
+        **kw,
+    ) -> "Converter._DatasetCtx":
+        loader = ShardedParquetLoader(
+            str(self.path),
+            batch_size=batch_size,
+            cur_shard=cur_shard,
+            shard_count=shard_count,
+            num_epochs=num_epochs,
+            workers=workers_count,
+            device=device,
+            **kw,
+        )
+        return Converter._DatasetCtx(loader)
+
+    def delete(self) -> None:
+        import shutil
+
+        if self._owned and self.path.exists():
+            shutil.rmtree(self.path)
+
+
+def make_converter(
+    source,
+    cache_dir: Optional[str] = None,
+    row_group_rows: int = 64,
+) -> Converter:
+    """Build a converter from a pyarrow Table (materialised to a Parquet cache
+    dir, like ``make_spark_converter``) or an existing Parquet dataset path."""
+    if isinstance(source, (str, Path)):
+        return Converter(Path(source), owned=False)
+    if not isinstance(source, pa.Table):
+        raise TypeError(f"unsupported source {type(source)}")
+    cache_root = Path(cache_dir or (Path(current_setup().root) / "cache"))
+    dst = cache_root / f"converter-{uuid.uuid4().hex[:12]}"
+    dst.mkdir(parents=True, exist_ok=True)
+    pq.write_table(source, dst / "part-00000.parquet", row_group_size=row_group_rows, compression="NONE")
+    return Converter(dst, owned=True)

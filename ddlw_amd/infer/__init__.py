@@ -1,0 +1,3 @@
+from .pyfunc import PythonModel, PyFuncModel, log_model, load_model, predict_udf
+
+__all__ = ["PythonModel", "PyFuncModel", "log_model", "load_model", "predict_udf"]

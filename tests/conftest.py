@@ -1,0 +1,46 @@
+import os
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO_ROOT = Path(__file__).resolve().parent.parent
+if str(REPO_ROOT) not in sys.path:
+    sys.path.insert(0, str(REPO_ROOT))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: requires an MI355X GPU (run on a gpurun box)")
+
+
+def pytest_collection_modifyitems(config, items):
+    import torch
+
+    if torch.cuda.is_available():
+        return
+    skip_gpu = pytest.mark.skip(reason="no GPU in this container")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip_gpu)
+
+
+@pytest.fixture()
+def ddlw_home(tmp_path, monkeypatch):
+    """Isolated DDLW_HOME + tracking store per test."""
+    home = tmp_path / "ddlw_home"
+    monkeypatch.setenv("DDLW_HOME", str(home))
+    monkeypatch.delenv("DDLW_TRACKING_URI", raising=False)
+    monkeypatch.delenv("DDLW_PARENT_RUN_ID", raising=False)
+    import ddlw_amd.core.config as config
+    import ddlw_amd.core.tracking as tracking
+
+    config._SETUP = None
+    tracking._tracking_uri = None
+    tracking._active_experiment_id = None
+    tracking._run_stack.clear()
+    config.setup(root=str(home))
+    yield home
+    tracking._run_stack.clear()
+    config._SETUP = None
+    tracking._tracking_uri = None
+    tracking._active_experiment_id = None

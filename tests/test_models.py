@@ -1,0 +1,53 @@
+"""Model-family tests: shapes, frozen-base semantics, save/load round trip."""
+import torch
+
+from ddlw_amd.core import tracking
+from ddlw_amd.core.model_io import load_model, log_model
+from ddlw_amd.models import build_model, build_resnet50, build_small_cnn
+
+
+def test_small_cnn_shapes():
+    m = build_small_cnn(num_classes=5)
+    y = m(torch.randn(2, 3, 64, 64))
+    assert y.shape == (2, 5)
+
+
+def test_resnet50_shapes():
+    m = build_resnet50(num_classes=7)
+    y = m(torch.randn(2, 3, 224, 224))
+    assert y.shape == (2, 7)
+    n_params = sum(p.numel() for p in m.parameters())
+    # ResNet-50 @1000 classes has 25.56M; at 7 classes ~23.5-24M
+    assert 20e6 < n_params < 30e6
+
+
+def test_transfer_model_frozen_base():
+    m = build_model(num_classes=5, dropout=0.5)
+    head_params = [p for p in m.parameters() if p.requires_grad]
+    # only the Dense head trains (reference P1/02:167-176)
+    assert sum(p.numel() for p in head_params) == 1280 * 5 + 5
+    m.train()
+    # frozen BN stays in eval mode (quirk 5)
+    assert not m.base.base.training
+    y = m(torch.randn(2, 3, 64, 64))
+    assert y.shape == (2, 5)
+    # backward only touches the head
+    y.sum().backward()
+    assert m.classifier.weight.grad is not None
+
+
+def test_model_save_load_roundtrip(ddlw_home):
+    tracking.set_experiment("models")
+    m = build_small_cnn(num_classes=4)
+    with tracking.start_run() as run:
+        uri = log_model(m, "model")
+    m2 = load_model(uri)
+    x = torch.randn(3, 3, 64, 64)
+    m.eval(), m2.eval()
+    assert torch.allclose(m(x), m2(x), atol=1e-6)
+    # also via the registry stage URI
+    tracking.register_model(uri, "cnn")
+    tracking.transition_model_version_stage("cnn", 1, "Production")
+    m3 = load_model("models:/cnn/production")
+    m3.eval()
+    assert torch.allclose(m(x), m3(x), atol=1e-6)

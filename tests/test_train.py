@@ -1,0 +1,103 @@
+"""Train-facade tests: fit/evaluate contract, callbacks, autolog."""
+import torch
+
+from ddlw_amd.core import tracking
+from ddlw_amd.models import build_small_cnn
+from ddlw_amd.train import (
+    EarlyStopping,
+    History,
+    LearningRateWarmupCallback,
+    Model,
+    ModelCheckpoint,
+    ReduceLROnPlateau,
+    autolog,
+)
+
+
+def _toy_data(n=32, bs=8, num_classes=3, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    xs = torch.randn(n, 3, 16, 16, generator=g)
+    ys = torch.randint(0, num_classes, (n,), generator=g)
+    return [(xs[i : i + bs], ys[i : i + bs]) for i in range(0, n, bs)]
+
+
+def test_fit_history_contract(ddlw_home):
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("Adam", learning_rate=1e-3)
+    hist = m.fit(_toy_data(), epochs=2, validation_data=_toy_data(seed=1), verbose=0)
+    assert isinstance(hist, History)
+    for key in ("loss", "accuracy", "val_loss", "val_accuracy"):
+        assert len(hist.history[key]) == 2
+
+
+def test_evaluate_returns_loss_and_metrics(ddlw_home):
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.1)
+    out = m.evaluate(_toy_data(), steps=2)
+    assert len(out) == 2  # [loss, accuracy]
+    d = m.evaluate(_toy_data(), steps=2, return_dict=True)
+    assert set(d) == {"loss", "accuracy"}
+
+
+def test_early_stopping_stops(ddlw_home):
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.0)
+    # lr=0 -> no improvement -> stops after patience+1 epochs without progress
+    hist = m.fit(
+        _toy_data(),
+        epochs=10,
+        validation_data=_toy_data(seed=1),
+        callbacks=[EarlyStopping(monitor="val_loss", min_delta=1e-9, patience=1)],
+        verbose=0,
+    )
+    assert len(hist.history["loss"]) < 10
+
+
+def test_reduce_lr_on_plateau(ddlw_home):
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=1.0)
+    cb = ReduceLROnPlateau(monitor="val_loss", factor=0.5, patience=0)
+    m.fit(
+        _toy_data(),
+        epochs=3,
+        validation_data=_toy_data(seed=1),
+        callbacks=[cb],
+        verbose=0,
+    )
+    assert m.optimizer.param_groups[0]["lr"] < 1.0
+
+
+def test_lr_warmup_ramps(ddlw_home):
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.8)
+    lrs = []
+
+    class Probe(LearningRateWarmupCallback):
+        def on_batch_begin(self, batch, logs=None):
+            super().on_batch_begin(batch, logs)
+            lrs.append(self.model.optimizer.param_groups[0]["lr"])
+
+    m.fit(_toy_data(), steps_per_epoch=4, epochs=2, callbacks=[Probe(warmup_epochs=2)], verbose=0)
+    assert lrs[-1] >= lrs[0]
+    assert abs(lrs[-1] - 0.8) < 1e-6
+
+
+def test_checkpoint_naming(ddlw_home, tmp_path):
+    ckpt = tmp_path / "trial" / "checkpoint-{epoch}.ckpt"
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.1)
+    m.fit(_toy_data(), epochs=2, callbacks=[ModelCheckpoint(str(ckpt))], verbose=0)
+    # reference layout: checkpoint-{epoch}.ckpt, weights only (P2/02:206-211)
+    assert (tmp_path / "trial" / "checkpoint-1.ckpt").exists()
+    assert (tmp_path / "trial" / "checkpoint-2.ckpt").exists()
+    sd = torch.load(tmp_path / "trial" / "checkpoint-2.ckpt", weights_only=True)
+    assert "classifier.weight" in sd
+
+
+def test_autolog(ddlw_home):
+    tracking.set_experiment("autolog")
+    autolog(True)
+    try:
+        with tracking.start_run() as run:
+            m = Model(build_small_cnn(16, 16, num_classes=3)).compile("Adam", learning_rate=1e-3)
+            m.fit(_toy_data(), epochs=1, verbose=0)
+        r = tracking.get_run(run.run_id)
+        assert "loss" in r.metrics()
+        assert r.params()["optimizer_name"] == "Adam"
+        assert (r.dir / "artifacts" / "model" / "state_dict.pt").exists()
+    finally:
+        autolog(False)
