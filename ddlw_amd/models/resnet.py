@@ -4,10 +4,13 @@ Written from scratch (no torchvision in this environment). Standard v1.5
 bottleneck architecture (stride-2 in the 3x3 conv of downsampling blocks),
 because that is what "ResNet-50 images/sec" conventionally measures.
 
-The module is built from ``ddlw_amd.ops`` layers, which dispatch to the
-hand-written CDNA4 HIP kernels when running on a gfx950 device and fall back
-to stock PyTorch ops on CPU (the CPU path doubles as the numerics oracle for
-kernel parity tests).
+Built from ``ddlw_amd.ops.layers`` blocks: BatchNorm+ReLU (and the block-end
+BatchNorm+residual-add+ReLU) are ONE fused hand-written CDNA4 HIP kernel each
+way when running bf16/channels_last on gfx950, as are the stem maxpool and
+the global average pool; on CPU the same layers fall back to stock PyTorch
+fp32 ops, which doubles as the numerics oracle for the kernel parity tests
+(SURVEY.md §4 item 2). Convolutions use MIOpen or the ddlw MFMA
+implicit-GEMM kernels depending on the ops dispatch setting.
 """
 from __future__ import annotations
 
@@ -17,6 +20,8 @@ import torch
 import torch.nn as nn
 
 from ..core.model_io import tag_model
+from ..ops.layers import BatchNormAct2d, GlobalAvgPool2d, MaxPool3x3s2
+from ..ops.conv import Conv2d
 
 
 class Bottleneck(nn.Module):
@@ -24,25 +29,31 @@ class Bottleneck(nn.Module):
 
     def __init__(self, in_ch: int, mid_ch: int, stride: int = 1, downsample: nn.Module = None):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_ch, mid_ch, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(mid_ch)
-        self.conv2 = nn.Conv2d(mid_ch, mid_ch, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(mid_ch)
-        self.conv3 = nn.Conv2d(mid_ch, mid_ch * self.expansion, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(mid_ch * self.expansion)
-        self.relu = nn.ReLU(inplace=True)
+        self.conv1 = Conv2d(in_ch, mid_ch, 1, bias=False)
+        self.bn1 = BatchNormAct2d(mid_ch, relu=True)
+        self.conv2 = Conv2d(mid_ch, mid_ch, 3, stride=stride, padding=1, bias=False)
+        self.bn2 = BatchNormAct2d(mid_ch, relu=True)
+        self.conv3 = Conv2d(mid_ch, mid_ch * self.expansion, 1, bias=False)
+        # bn3 + residual add + relu fused into one kernel
+        self.bn3 = BatchNormAct2d(mid_ch * self.expansion, relu=True)
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        out = out + identity
-        return self.relu(out)
+        identity = self.downsample(x) if self.downsample is not None else x
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=identity)
+
+
+class Downsample(nn.Module):
+    def __init__(self, in_ch: int, out_ch: int, stride: int):
+        super().__init__()
+        self.conv = Conv2d(in_ch, out_ch, 1, stride=stride, bias=False)
+        self.bn = BatchNormAct2d(out_ch, relu=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.bn(self.conv(x))
 
 
 class ResNet50(nn.Module):
@@ -50,15 +61,14 @@ class ResNet50(nn.Module):
         super().__init__()
         layers = [3, 4, 6, 3]
         self.in_ch = 64
-        self.conv1 = nn.Conv2d(channels, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
-        self.relu = nn.ReLU(inplace=True)
-        self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
+        self.conv1 = Conv2d(channels, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = BatchNormAct2d(64, relu=True)
+        self.maxpool = MaxPool3x3s2()
         self.layer1 = self._make_layer(64, layers[0], stride=1)
         self.layer2 = self._make_layer(128, layers[1], stride=2)
         self.layer3 = self._make_layer(256, layers[2], stride=2)
         self.layer4 = self._make_layer(512, layers[3], stride=2)
-        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.avgpool = GlobalAvgPool2d()
         self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
         self._init_weights()
 
@@ -66,10 +76,7 @@ class ResNet50(nn.Module):
         downsample = None
         out_ch = mid_ch * Bottleneck.expansion
         if stride != 1 or self.in_ch != out_ch:
-            downsample = nn.Sequential(
-                nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(out_ch),
-            )
+            downsample = Downsample(self.in_ch, out_ch, stride)
         layer: List[nn.Module] = [Bottleneck(self.in_ch, mid_ch, stride, downsample)]
         self.in_ch = out_ch
         for _ in range(1, blocks):
@@ -78,9 +85,9 @@ class ResNet50(nn.Module):
 
     def _init_weights(self) -> None:
         for m in self.modules():
-            if isinstance(m, nn.Conv2d):
+            if isinstance(m, (nn.Conv2d, Conv2d)):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
+            elif isinstance(m, BatchNormAct2d):
                 nn.init.ones_(m.weight)
                 nn.init.zeros_(m.bias)
         # zero-init the last BN of each block (standard ResNet-50 recipe)
@@ -89,13 +96,13 @@ class ResNet50(nn.Module):
                 nn.init.zeros_(m.bn3.weight)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = self.bn1(self.conv1(x))
         x = self.maxpool(x)
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)
         x = self.layer4(x)
-        x = self.avgpool(x).flatten(1)
+        x = self.avgpool(x)
         return self.fc(x)
 
 

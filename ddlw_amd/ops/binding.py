@@ -1,0 +1,208 @@
+"""ctypes bindings for libddlw_kernels.so (typed, tensor-level wrappers).
+
+All activation tensors must be channels_last (NHWC memory) bf16; statistics
+and parameters fp32. Every call launches onto torch's current HIP stream, so
+the kernels are stream-ordered with PyTorch ops and hipGraph-capturable.
+"""
+from __future__ import annotations
+
+import ctypes
+from typing import Optional
+
+import torch
+
+from .runtime import check, current_stream_ptr, require_lib
+
+
+def _p(t: Optional[torch.Tensor]):
+    return ctypes.c_void_p(0 if t is None else t.data_ptr())
+
+
+def _nhwc_ptr(t: torch.Tensor):
+    """Data pointer of a channels_last activation (N,C,H,W logical)."""
+    if t.dim() == 4:
+        assert t.is_contiguous(memory_format=torch.channels_last), "need channels_last"
+    else:
+        assert t.is_contiguous()
+    assert t.dtype == torch.bfloat16, f"need bf16, got {t.dtype}"
+    return _p(t)
+
+
+def _rows_c(t: torch.Tensor):
+    if t.dim() == 4:
+        n, c, h, w = t.shape
+        return n * h * w, c
+    rows, c = t.shape[0], t.shape[-1]
+    return rows, c
+
+
+def supported_channels(c: int) -> bool:
+    return c % 8 == 0
+
+
+def bn_stats(x: torch.Tensor, eps: float, momentum: float,
+             running_mean: Optional[torch.Tensor], running_var: Optional[torch.Tensor]):
+    lib = require_lib()
+    rows, C = _rows_c(x)
+    dev = x.device
+    sum_ = torch.zeros(C, dtype=torch.float32, device=dev)
+    sumsq = torch.zeros(C, dtype=torch.float32, device=dev)
+    mean = torch.empty(C, dtype=torch.float32, device=dev)
+    rstd = torch.empty(C, dtype=torch.float32, device=dev)
+    s = current_stream_ptr()
+    check(lib.ddlw_bn_stats(_nhwc_ptr(x), _p(sum_), _p(sumsq),
+                            ctypes.c_long(rows), ctypes.c_int(C),
+                            ctypes.c_void_p(s)), "bn_stats")
+    check(lib.ddlw_bn_finalize(_p(sum_), _p(sumsq), _p(mean), _p(rstd),
+                               _p(running_mean), _p(running_var),
+                               ctypes.c_long(rows), ctypes.c_int(C),
+                               ctypes.c_float(eps), ctypes.c_float(momentum),
+                               ctypes.c_void_p(s)), "bn_finalize")
+    return mean, rstd
+
+
+def bn_apply(x: torch.Tensor, res: Optional[torch.Tensor], mean, rstd, gamma, beta,
+             relu: bool) -> torch.Tensor:
+    lib = require_lib()
+    rows, C = _rows_c(x)
+    y = torch.empty_like(x)
+    check(
+        lib.ddlw_bn_apply(
+            _nhwc_ptr(x), _p(res), _nhwc_ptr(y), _p(mean), _p(rstd), _p(gamma),
+            _p(beta), ctypes.c_long(rows), ctypes.c_int(C),
+            ctypes.c_int(1 if relu else 0), ctypes.c_void_p(current_stream_ptr())
+        ),
+        "bn_apply",
+    )
+    return y
+
+
+def bn_bwd_reduce(dy, y, x, mean, rstd, relu: bool):
+    lib = require_lib()
+    rows, C = _rows_c(x)
+    dev = x.device
+    dbeta = torch.zeros(C, dtype=torch.float32, device=dev)
+    dgamma = torch.zeros(C, dtype=torch.float32, device=dev)
+    check(
+        lib.ddlw_bn_bwd_reduce(
+            _nhwc_ptr(dy), _p(y), _nhwc_ptr(x), _p(mean), _p(rstd), _p(dbeta),
+            _p(dgamma), ctypes.c_long(rows), ctypes.c_int(C),
+            ctypes.c_int(1 if relu else 0), ctypes.c_void_p(current_stream_ptr())
+        ),
+        "bn_bwd_reduce",
+    )
+    return dbeta, dgamma
+
+
+def bn_bwd_dx(dy, y, x, mean, rstd, gamma, dbeta, dgamma, relu: bool,
+              want_dres: bool):
+    lib = require_lib()
+    rows, C = _rows_c(x)
+    dx = torch.empty_like(x)
+    dres = torch.empty_like(x) if want_dres else None
+    check(
+        lib.ddlw_bn_bwd_dx(
+            _nhwc_ptr(dy), _p(y), _nhwc_ptr(x), _p(mean), _p(rstd), _p(gamma),
+            _p(dbeta), _p(dgamma), _p(dx), _p(dres), ctypes.c_long(rows),
+            ctypes.c_int(C), ctypes.c_int(1 if relu else 0),
+            ctypes.c_void_p(current_stream_ptr())
+        ),
+        "bn_bwd_dx",
+    )
+    return dx, dres
+
+
+def maxpool3x3s2_fwd(x: torch.Tensor):
+    lib = require_lib()
+    n, c, h, w = x.shape
+    ho, wo = (h + 1) // 2, (w + 1) // 2
+    y = torch.empty((n, c, ho, wo), dtype=x.dtype, device=x.device).to(
+        memory_format=torch.channels_last
+    )
+    argmax = torch.empty(n * ho * wo * c, dtype=torch.uint8, device=x.device)
+    check(
+        lib.ddlw_maxpool3x3s2_fwd(
+            _nhwc_ptr(x), _nhwc_ptr(y), _p(argmax), n, h, w, c, ho, wo,
+            ctypes.c_void_p(current_stream_ptr())
+        ),
+        "maxpool_fwd",
+    )
+    return y, argmax
+
+
+def maxpool3x3s2_bwd(dy: torch.Tensor, argmax: torch.Tensor, in_shape):
+    lib = require_lib()
+    n, c, h, w = in_shape
+    ho, wo = dy.shape[2], dy.shape[3]
+    dx = torch.empty((n, c, h, w), dtype=dy.dtype, device=dy.device).to(
+        memory_format=torch.channels_last
+    )
+    check(
+        lib.ddlw_maxpool3x3s2_bwd(
+            _nhwc_ptr(dy), _p(argmax), _nhwc_ptr(dx), n, h, w, c, ho, wo,
+            ctypes.c_void_p(current_stream_ptr())
+        ),
+        "maxpool_bwd",
+    )
+    return dx
+
+
+def gap_fwd(x: torch.Tensor) -> torch.Tensor:
+    lib = require_lib()
+    n, c, h, w = x.shape
+    y = torch.empty((n, c), dtype=x.dtype, device=x.device)
+    check(lib.ddlw_gap_fwd(_nhwc_ptr(x), _p(y), n, h * w, c,
+                           ctypes.c_void_p(current_stream_ptr())), "gap_fwd")
+    return y
+
+
+def gap_bwd(dy: torch.Tensor, in_shape) -> torch.Tensor:
+    lib = require_lib()
+    n, c, h, w = in_shape
+    dx = torch.empty((n, c, h, w), dtype=dy.dtype, device=dy.device).to(
+        memory_format=torch.channels_last
+    )
+    check(lib.ddlw_gap_bwd(_p(dy.contiguous()), _nhwc_ptr(dx), n, h * w, c,
+                           ctypes.c_void_p(current_stream_ptr())), "gap_bwd")
+    return dx
+
+
+def softmax_ce(logits: torch.Tensor, labels: torch.Tensor, grad_scale: float):
+    """Returns (mean loss tensor [1], dlogits). logits fp32 [B,K]."""
+    lib = require_lib()
+    B, K = logits.shape
+    logits = logits.contiguous()
+    dlogits = torch.empty_like(logits)
+    loss_sum = torch.zeros(1, dtype=torch.float32, device=logits.device)
+    check(
+        lib.ddlw_softmax_ce(
+            _p(logits), _p(labels.contiguous()), _p(dlogits), _p(loss_sum), B, K,
+            ctypes.c_float(grad_scale), ctypes.c_void_p(current_stream_ptr())
+        ),
+        "softmax_ce",
+    )
+    return loss_sum / B, dlogits
+
+
+def fused_sgd(chunk_desc: torch.Tensor, nchunks: int, max_numel: int, lr: float,
+              momentum: float, weight_decay: float, first_step: bool):
+    """chunk_desc: int64 device tensor [nchunks, 5] of
+    (p_ptr, g_ptr, m_ptr, p_bf16_ptr_or_0, numel)."""
+    lib = require_lib()
+    check(
+        lib.ddlw_fused_sgd(
+            _p(chunk_desc), nchunks, ctypes.c_long(max_numel), ctypes.c_float(lr),
+            ctypes.c_float(momentum), ctypes.c_float(weight_decay),
+            ctypes.c_int(1 if first_step else 0),
+            ctypes.c_void_p(current_stream_ptr())
+        ),
+        "fused_sgd",
+    )
+
+
+def normalize_u8(x_u8: torch.Tensor, out_bf16: torch.Tensor):
+    lib = require_lib()
+    n = x_u8.numel()
+    assert n % 16 == 0
+    check(lib.ddlw_normalize_u8(_p(x_u8), _p(out_bf16), ctypes.c_long(n),
+                                ctypes.c_void_p(current_stream_ptr())), "normalize_u8")

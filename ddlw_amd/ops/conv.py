@@ -1,0 +1,45 @@
+"""Conv2d with dispatch between the ddlw MFMA implicit-GEMM HIP kernels and
+the library path (MIOpen via F.conv2d).
+
+The reference's convolutions live behind TF/Keras (kernels K1-K3 in
+SURVEY.md §2.4); here they are first-class. Dispatch policy via
+``DDLW_CONV`` env: ``auto`` (default — per-shape pick from the measured
+table), ``hip`` (force hand-written kernels), ``stock`` (force MIOpen).
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+def conv_mode() -> str:
+    return os.environ.get("DDLW_CONV", "auto")
+
+
+class Conv2d(nn.Conv2d):
+    """nn.Conv2d subclass so init/state_dict/bias handling are inherited;
+    forward dispatches per policy. HIP implicit-GEMM path: ops.conv_gemm."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        mode = conv_mode()
+        if (
+            mode != "stock"
+            and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and self.groups == 1
+        ):
+            from . import conv_gemm
+
+            if conv_gemm.available(self, x, mode):
+                return conv_gemm.conv2d(
+                    x, self.weight, self.bias, self.stride, self.padding
+                )
+        return F.conv2d(
+            x, self.weight.to(x.dtype),
+            None if self.bias is None else self.bias.to(x.dtype),
+            self.stride, self.padding, self.dilation, self.groups,
+        )

@@ -1,0 +1,649 @@
+// Fused memory-bound kernels for the ResNet-50 training step (SURVEY.md §2.4
+// kernel table K4-K12 + BASELINE "ResNet-50 additions" row), NHWC bf16.
+//
+// Design (MI355X): these ops are HBM-bound — the levers are vectorized 16-B
+// bf16x8 accesses, pass fusion (BN+ReLU+residual in one read/write of the
+// activation), and fp32 accumulation. Grids are capped ~2048 workgroups of
+// 256 threads with grid-stride loops (cdna_hip_programming.md Guideline 11).
+#include "common.h"
+
+static __thread char g_err[256];
+void ddlw_set_error(const char* msg) {
+  int i = 0;
+  for (; msg[i] && i < 255; ++i) g_err[i] = msg[i];
+  g_err[i] = 0;
+}
+DDLW_EXPORT const char* ddlw_last_error() { return g_err; }
+
+DDLW_EXPORT int ddlw_device_sync() {
+  hipError_t e = hipDeviceSynchronize();
+  if (e != hipSuccess) { ddlw_set_error(hipGetErrorString(e)); return 1; }
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// BatchNorm training statistics: per-channel sum / sumsq over rows = N*H*W.
+// x viewed as [rows][C]; C % 8 == 0 and C/8 is a power of two <= 256
+// (every ResNet-50 / MobileNetV2 channel count satisfies this; the generic
+// fallback tile loop covers other C).
+// Each thread owns 8 consecutive channels (one bf16x8 vector); thread groups
+// stack over rows; LDS tree-reduce across row groups; one fp32 atomicAdd per
+// channel into the zero-initialised global accumulators.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_bn_stats(
+    const bf16_t* __restrict__ x, float* __restrict__ sum, float* __restrict__ sumsq,
+    long rows, int C) {
+  const int vecC = C >> 3;
+  const int VPB = vecC < 256 ? vecC : 256;      // vectors per block
+  const int ROWG = 256 / VPB;                   // row groups per block (floor)
+  const int tid = threadIdx.x;
+  const int vec = (blockIdx.x * VPB) + (tid % VPB);
+  const int rowg = tid / VPB;
+  // inactive threads still reach every barrier (no early return!)
+  const bool active = (rowg < ROWG) && (vec < vecC);
+
+  float s[8] = {0}, q[8] = {0};
+  const long row0 = (long)blockIdx.y * ROWG + rowg;
+  const long rstride = (long)gridDim.y * ROWG;
+  if (active)
+    for (long r = row0; r < rows; r += rstride) {
+      bf16x8 v;
+      v.v = *reinterpret_cast<const uint4*>(x + r * C + (long)vec * 8);
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float f = b2f(v.h[k]);
+        s[k] += f;
+        q[k] += f * f;
+      }
+    }
+  // reduce across row groups through LDS
+  __shared__ float lds[256 * 8];
+  if (ROWG > 1) {
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) lds[tid * 8 + k] = s[k];
+    __syncthreads();
+    if (rowg == 0) {
+      for (int g = 1; g < ROWG; ++g)
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) s[k] += lds[(g * VPB + (tid % VPB)) * 8 + k];
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) lds[tid * 8 + k] = q[k];
+    __syncthreads();
+    if (rowg == 0) {
+      for (int g = 1; g < ROWG; ++g)
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) q[k] += lds[(g * VPB + (tid % VPB)) * 8 + k];
+    }
+  }
+  if (active && rowg == 0) {
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      atomicAdd(sum + vec * 8 + k, s[k]);
+      atomicAdd(sumsq + vec * 8 + k, q[k]);
+    }
+  }
+}
+
+// finalize: mean/rstd + running-stat update (train-mode BN, K4 fwd+bwd row)
+__global__ __launch_bounds__(256) void k_bn_finalize(
+    const float* __restrict__ sum, const float* __restrict__ sumsq,
+    float* __restrict__ mean, float* __restrict__ rstd,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    long rows, int C, float eps, float momentum) {
+  int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= C) return;
+  float m = sum[c] / (float)rows;
+  float var = fmaxf(sumsq[c] / (float)rows - m * m, 0.f);
+  mean[c] = m;
+  rstd[c] = rsqrtf(var + eps);
+  if (running_mean) {
+    float unbiased = var * (float)rows / (float)(rows > 1 ? rows - 1 : 1);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// BN apply (+ optional residual add, + optional ReLU), one fused pass:
+//   y = act((x - mean[c]) * rstd[c] * gamma[c] + beta[c] [+ res])
+// RELU: 0 = identity, 1 = relu. res may be null.
+// ---------------------------------------------------------------------------
+template <int RELU, bool HAS_RES>
+__global__ __launch_bounds__(256) void k_bn_apply(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ res,
+    bf16_t* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    long rows, int C) {
+  const int vecC = C >> 3;
+  const long nvec = rows * vecC;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    const int vc = (int)(i % vecC);
+    bf16x8 v, o, rv;
+    v.v = *reinterpret_cast<const uint4*>(x + i * 8);
+    if (HAS_RES) rv.v = *reinterpret_cast<const uint4*>(res + i * 8);
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const int c = vc * 8 + k;
+      float f = (b2f(v.h[k]) - mean[c]) * rstd[c] * gamma[c] + beta[c];
+      if (HAS_RES) f += b2f(rv.h[k]);
+      if (RELU) f = fmaxf(f, 0.f);
+      o.h[k] = f2b(f);
+    }
+    *reinterpret_cast<uint4*>(y + i * 8) = o.v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// BN backward reductions: with fused ReLU the incoming dy must be masked by
+// (y > 0) first; we take y (the saved activation output) and produce
+//   dbeta[c]  = sum(dy_m),  dgamma[c] = sum(dy_m * xhat)
+// Same thread geometry as k_bn_stats.
+// ---------------------------------------------------------------------------
+template <int RELU>
+__global__ __launch_bounds__(256) void k_bn_bwd_reduce(
+    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
+    const bf16_t* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ dbeta, float* __restrict__ dgamma,
+    long rows, int C) {
+  const int vecC = C >> 3;
+  const int VPB = vecC < 256 ? vecC : 256;
+  const int ROWG = 256 / VPB;
+  const int tid = threadIdx.x;
+  const int vec = (blockIdx.x * VPB) + (tid % VPB);
+  const int rowg = tid / VPB;
+  const bool active = (rowg < ROWG) && (vec < vecC);
+
+  float db[8] = {0}, dg[8] = {0};
+  float mn[8], rs[8];
+  if (active) {
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) { mn[k] = mean[vec * 8 + k]; rs[k] = rstd[vec * 8 + k]; }
+    const long row0 = (long)blockIdx.y * ROWG + rowg;
+    const long rstride = (long)gridDim.y * ROWG;
+    for (long r = row0; r < rows; r += rstride) {
+      const long base = r * C + (long)vec * 8;
+      bf16x8 vdy, vy, vx;
+      vdy.v = *reinterpret_cast<const uint4*>(dy + base);
+      if (RELU) vy.v = *reinterpret_cast<const uint4*>(y + base);
+      vx.v = *reinterpret_cast<const uint4*>(x + base);
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = b2f(vdy.h[k]);
+        if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
+        db[k] += g;
+        dg[k] += g * (b2f(vx.h[k]) - mn[k]) * rs[k];
+      }
+    }
+  }
+  __shared__ float lds[256 * 8];
+  if (ROWG > 1) {
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) lds[tid * 8 + k] = db[k];
+    __syncthreads();
+    if (rowg == 0)
+      for (int g = 1; g < ROWG; ++g)
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) db[k] += lds[(g * VPB + (tid % VPB)) * 8 + k];
+    __syncthreads();
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) lds[tid * 8 + k] = dg[k];
+    __syncthreads();
+    if (rowg == 0)
+      for (int g = 1; g < ROWG; ++g)
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) dg[k] += lds[(g * VPB + (tid % VPB)) * 8 + k];
+  }
+  if (active && rowg == 0) {
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      atomicAdd(dbeta + vec * 8 + k, db[k]);
+      atomicAdd(dgamma + vec * 8 + k, dg[k]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// BN backward dx (+ optional residual grad out):
+//   dy_m = RELU ? dy * (y > 0) : dy
+//   dx   = gamma*rstd * (dy_m - dbeta/M - xhat * dgamma/M)
+//   dres = dy_m (residual branch gets the masked upstream grad)
+// ---------------------------------------------------------------------------
+template <int RELU, bool HAS_RES>
+__global__ __launch_bounds__(256) void k_bn_bwd_dx(
+    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
+    const bf16_t* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma,
+    const float* __restrict__ dbeta, const float* __restrict__ dgamma,
+    bf16_t* __restrict__ dx, bf16_t* __restrict__ dres,
+    long rows, int C) {
+  const int vecC = C >> 3;
+  const long nvec = rows * vecC;
+  const float invM = 1.f / (float)rows;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    const int vc = (int)(i % vecC);
+    bf16x8 vdy, vy, vx, odx, odr;
+    vdy.v = *reinterpret_cast<const uint4*>(dy + i * 8);
+    if (RELU) vy.v = *reinterpret_cast<const uint4*>(y + i * 8);
+    vx.v = *reinterpret_cast<const uint4*>(x + i * 8);
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const int c = vc * 8 + k;
+      float g = b2f(vdy.h[k]);
+      if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
+      float xhat = (b2f(vx.h[k]) - mean[c]) * rstd[c];
+      float d = gamma[c] * rstd[c] * (g - dbeta[c] * invM - xhat * dgamma[c] * invM);
+      odx.h[k] = f2b(d);
+      if (HAS_RES) odr.h[k] = f2b(g);
+    }
+    *reinterpret_cast<uint4*>(dx + i * 8) = odx.v;
+    if (HAS_RES) *reinterpret_cast<uint4*>(dres + i * 8) = odr.v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MaxPool 3x3 stride 2 pad 1, NHWC (the ResNet stem pool; K "maxpool" row).
+// fwd records the argmax window slot (0..8) for an atomics-free backward.
+// One thread per 8-channel vector of one output pixel.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_maxpool3x3s2_fwd(
+    const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+    unsigned char* __restrict__ argmax,
+    int N, int H, int W, int C, int Ho, int Wo) {
+  const int vecC = C >> 3;
+  const long total = (long)N * Ho * Wo * vecC;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int vc = (int)(i % vecC);
+    long t = i / vecC;
+    int wo = (int)(t % Wo); t /= Wo;
+    int ho = (int)(t % Ho); t /= Ho;
+    int n = (int)t;
+    float best[8];
+    int bidx[8];
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) { best[k] = -3.4e38f; bidx[k] = 0; }
+    #pragma unroll
+    for (int kh = 0; kh < 3; ++kh) {
+      int h = ho * 2 - 1 + kh;
+      if (h < 0 || h >= H) continue;
+      #pragma unroll
+      for (int kw = 0; kw < 3; ++kw) {
+        int w = wo * 2 - 1 + kw;
+        if (w < 0 || w >= W) continue;
+        bf16x8 v;
+        v.v = *reinterpret_cast<const uint4*>(
+            x + (((long)n * H + h) * W + w) * C + (long)vc * 8);
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          float f = b2f(v.h[k]);
+          if (f > best[k]) { best[k] = f; bidx[k] = kh * 3 + kw; }
+        }
+      }
+    }
+    bf16x8 o;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o.h[k] = f2b(best[k]);
+    *reinterpret_cast<uint4*>(y + i * 8) = o.v;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) argmax[i * 8 + k] = (unsigned char)bidx[k];
+  }
+}
+
+// backward: gather — for each input element, sum dy over the <=4 output
+// windows that could have selected it, checking the recorded argmax.
+__global__ __launch_bounds__(256) void k_maxpool3x3s2_bwd(
+    const bf16_t* __restrict__ dy, const unsigned char* __restrict__ argmax,
+    bf16_t* __restrict__ dx,
+    int N, int H, int W, int C, int Ho, int Wo) {
+  const int vecC = C >> 3;
+  const long total = (long)N * H * W * vecC;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int vc = (int)(i % vecC);
+    long t = i / vecC;
+    int w = (int)(t % W); t /= W;
+    int h = (int)(t % H); t /= H;
+    int n = (int)t;
+    float acc[8] = {0};
+    // output windows covering (h, w): ho*2-1 <= h <= ho*2+1
+    int ho_lo = (h - 1 + 1) / 2, ho_hi = (h + 1) / 2;  // ceil((h-1)/2), floor((h+1)/2)
+    if (h == 0) ho_lo = 0;
+    int wo_lo = (w - 1 + 1) / 2, wo_hi = (w + 1) / 2;
+    if (w == 0) wo_lo = 0;
+    for (int ho = ho_lo; ho <= ho_hi && ho < Ho; ++ho) {
+      int kh = h - (ho * 2 - 1);
+      if (kh < 0 || kh > 2) continue;
+      for (int wo = wo_lo; wo <= wo_hi && wo < Wo; ++wo) {
+        int kw = w - (wo * 2 - 1);
+        if (kw < 0 || kw > 2) continue;
+        long obase = (((long)n * Ho + ho) * Wo + wo) * C + (long)vc * 8;
+        bf16x8 g;
+        g.v = *reinterpret_cast<const uint4*>(dy + obase);
+        unsigned char slot = (unsigned char)(kh * 3 + kw);
+        #pragma unroll
+        for (int k = 0; k < 8; ++k)
+          if (argmax[obase + k] == slot) acc[k] += b2f(g.h[k]);
+      }
+    }
+    bf16x8 o;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o.h[k] = f2b(acc[k]);
+    *reinterpret_cast<uint4*>(dx + i * 8) = o.v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Global average pooling NHWC: y[n,c] = mean_hw x[n,h,w,c]  (K6).
+// One block per image n; threads own channel vectors; loop over H*W rows.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_gap_fwd(
+    const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+    int HW, int C) {
+  const int vecC = C >> 3;
+  const int n = blockIdx.x;
+  const float inv = 1.f / (float)HW;
+  for (int vc = threadIdx.x; vc < vecC; vc += blockDim.x) {
+    float acc[8] = {0};
+    const bf16_t* base = x + (long)n * HW * C + (long)vc * 8;
+    for (int r = 0; r < HW; ++r) {
+      bf16x8 v;
+      v.v = *reinterpret_cast<const uint4*>(base + (long)r * C);
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) acc[k] += b2f(v.h[k]);
+    }
+    bf16x8 o;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o.h[k] = f2b(acc[k] * inv);
+    *reinterpret_cast<uint4*>(y + (long)n * C + (long)vc * 8) = o.v;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_gap_bwd(
+    const bf16_t* __restrict__ dy, bf16_t* __restrict__ dx,
+    int HW, int C, long total /* N*HW*vecC */) {
+  const int vecC = C >> 3;
+  const float inv = 1.f / (float)HW;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int vc = (int)(i % vecC);
+    long n = i / ((long)vecC * HW);
+    bf16x8 g, o;
+    g.v = *reinterpret_cast<const uint4*>(dy + n * C + (long)vc * 8);
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o.h[k] = f2b(b2f(g.h[k]) * inv);
+    *reinterpret_cast<uint4*>(dx + i * 8) = o.v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused softmax + sparse cross-entropy, fwd+bwd in one pass (K9):
+// logits fp32 [B, K] (the classifier head outputs fp32), labels i64.
+// One wave per row: online max + sumexp via shuffle reduction, then
+// dlogits = (softmax - onehot) * grad_scale, loss accumulated per row.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_softmax_ce(
+    const float* __restrict__ logits, const long* __restrict__ labels,
+    float* __restrict__ dlogits, float* __restrict__ loss_sum,
+    int B, int K, float grad_scale) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  const int nwaves = (gridDim.x * blockDim.x) >> 6;
+  for (int b = wave; b < B; b += nwaves) {
+    const float* row = logits + (long)b * K;
+    float mx = -3.4e38f;
+    for (int k = lane; k < K; k += 64) mx = fmaxf(mx, row[k]);
+    mx = warp_reduce_max(mx);
+    mx = __shfl(mx, 0, 64);
+    float se = 0.f;
+    for (int k = lane; k < K; k += 64) se += __expf(row[k] - mx);
+    se = warp_reduce_sum(se);
+    se = __shfl(se, 0, 64);
+    const float inv_se = 1.f / se;
+    const long lab = labels[b];
+    for (int k = lane; k < K; k += 64) {
+      float p = __expf(row[k] - mx) * inv_se;
+      dlogits[(long)b * K + k] = (p - (k == (int)lab ? 1.f : 0.f)) * grad_scale;
+    }
+    if (lane == 0) {
+      float logp = row[lab] - mx - __logf(se);
+      atomicAdd(loss_sum, -logp);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused SGD (momentum + weight decay + nesterov-free), multi-tensor (K10 /
+// "fused SGD step" in the BASELINE north star). Descriptor array of
+// {param fp32, grad fp32, momentum fp32, numel} chunks lives in device mem.
+// Optional bf16 shadow copy of the weights (model weights in bf16 while the
+// master stays fp32).
+// ---------------------------------------------------------------------------
+struct SgdChunk {
+  float* p;
+  const float* g;
+  float* m;
+  bf16_t* p_bf16;  // nullable
+  long n;
+};
+
+__global__ __launch_bounds__(256) void k_fused_sgd(
+    const SgdChunk* __restrict__ chunks, int nchunks,
+    float lr, float momentum, float weight_decay, int first_step) {
+  for (int ci = blockIdx.y; ci < nchunks; ci += gridDim.y) {
+    SgdChunk ch = chunks[ci];
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < ch.n;
+         i += (long)gridDim.x * blockDim.x) {
+      float g = ch.g[i] + weight_decay * ch.p[i];
+      float v = first_step ? g : momentum * ch.m[i] + g;
+      ch.m[i] = v;
+      float p = ch.p[i] - lr * v;
+      ch.p[i] = p;
+      if (ch.p_bf16) ch.p_bf16[i] = f2b(p);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Input normalize: uint8 NHWC -> bf16 NHWC, x/127.5 - 1 (the MobileNetV2
+// preprocess_input transform, reference P1/02:126), fused with the H2D'd
+// uint8 batch so the fp32 intermediate never exists.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_normalize_u8(
+    const unsigned char* __restrict__ x, bf16_t* __restrict__ y, long n16 /* n/16 */) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n16;
+       i += (long)gridDim.x * blockDim.x) {
+    uint4 v = *reinterpret_cast<const uint4*>(x + i * 16);
+    const unsigned char* b = reinterpret_cast<const unsigned char*>(&v);
+    bf16x8 o0, o1;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o0.h[k] = f2b((float)b[k] * (1.f / 127.5f) - 1.f);
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o1.h[k] = f2b((float)b[8 + k] * (1.f / 127.5f) - 1.f);
+    *reinterpret_cast<uint4*>(y + i * 16) = o0.v;
+    *reinterpret_cast<uint4*>(y + i * 16 + 8) = o1.v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+static inline int grid_1d(long work, int block = 256, int cap = 2048) {
+  long g = (work + block - 1) / block;
+  return (int)(g < cap ? (g > 0 ? g : 1) : cap);
+}
+
+DDLW_EXPORT int ddlw_bn_stats(const void* x, void* sum, void* sumsq,
+                              long rows, int C, void* stream) {
+  int vecC = C >> 3;
+  int VPB = vecC < 256 ? vecC : 256;
+  int ROWG = 256 / VPB;
+  dim3 grid((vecC + VPB - 1) / VPB, (int)min((rows + ROWG - 1) / ROWG, (long)1024));
+  hipLaunchKernelGGL(k_bn_stats, grid, dim3(256), 0, (hipStream_t)stream,
+                     (const bf16_t*)x, (float*)sum, (float*)sumsq, rows, C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_bn_finalize(const void* sum, const void* sumsq, void* mean,
+                                 void* rstd, void* rmean, void* rvar, long rows,
+                                 int C, float eps, float momentum, void* stream) {
+  hipLaunchKernelGGL(k_bn_finalize, dim3((C + 255) / 256), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)sum, (const float*)sumsq,
+                     (float*)mean, (float*)rstd, (float*)rmean, (float*)rvar,
+                     rows, C, eps, momentum);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_bn_apply(const void* x, const void* res, void* y,
+                              const void* mean, const void* rstd,
+                              const void* gamma, const void* beta, long rows,
+                              int C, int relu, void* stream) {
+  long nvec = rows * (C >> 3);
+  dim3 grid(grid_1d(nvec));
+  if (relu && res)
+    hipLaunchKernelGGL((k_bn_apply<1, true>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)x, (const bf16_t*)res, (bf16_t*)y,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)beta, rows, C);
+  else if (relu)
+    hipLaunchKernelGGL((k_bn_apply<1, false>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)x, nullptr, (bf16_t*)y,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)beta, rows, C);
+  else if (res)
+    hipLaunchKernelGGL((k_bn_apply<0, true>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)x, (const bf16_t*)res, (bf16_t*)y,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)beta, rows, C);
+  else
+    hipLaunchKernelGGL((k_bn_apply<0, false>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)x, nullptr, (bf16_t*)y,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)beta, rows, C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* y, const void* x,
+                                   const void* mean, const void* rstd,
+                                   void* dbeta, void* dgamma, long rows, int C,
+                                   int relu, void* stream) {
+  int vecC = C >> 3;
+  int VPB = vecC < 256 ? vecC : 256;
+  int ROWG = 256 / VPB;
+  dim3 grid((vecC + VPB - 1) / VPB, (int)min((rows + ROWG - 1) / ROWG, (long)1024));
+  if (relu)
+    hipLaunchKernelGGL((k_bn_bwd_reduce<1>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const float*)mean, (const float*)rstd, (float*)dbeta,
+                       (float*)dgamma, rows, C);
+  else
+    hipLaunchKernelGGL((k_bn_bwd_reduce<0>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const float*)mean, (const float*)rstd, (float*)dbeta,
+                       (float*)dgamma, rows, C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_bn_bwd_dx(const void* dy, const void* y, const void* x,
+                               const void* mean, const void* rstd,
+                               const void* gamma, const void* dbeta,
+                               const void* dgamma, void* dx, void* dres,
+                               long rows, int C, int relu, void* stream) {
+  long nvec = rows * (C >> 3);
+  dim3 grid(grid_1d(nvec));
+  if (relu && dres)
+    hipLaunchKernelGGL((k_bn_bwd_dx<1, true>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
+                       (bf16_t*)dres, rows, C);
+  else if (relu)
+    hipLaunchKernelGGL((k_bn_bwd_dx<1, false>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
+                       nullptr, rows, C);
+  else if (dres)
+    hipLaunchKernelGGL((k_bn_bwd_dx<0, true>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
+                       (bf16_t*)dres, rows, C);
+  else
+    hipLaunchKernelGGL((k_bn_bwd_dx<0, false>), grid, dim3(256), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const float*)mean, (const float*)rstd, (const float*)gamma,
+                       (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
+                       nullptr, rows, C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_maxpool3x3s2_fwd(const void* x, void* y, void* argmax,
+                                      int N, int H, int W, int C, int Ho, int Wo,
+                                      void* stream) {
+  long total = (long)N * Ho * Wo * (C >> 3);
+  hipLaunchKernelGGL(k_maxpool3x3s2_fwd, dim3(grid_1d(total)), dim3(256), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)y,
+                     (unsigned char*)argmax, N, H, W, C, Ho, Wo);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_maxpool3x3s2_bwd(const void* dy, const void* argmax, void* dx,
+                                      int N, int H, int W, int C, int Ho, int Wo,
+                                      void* stream) {
+  long total = (long)N * H * W * (C >> 3);
+  hipLaunchKernelGGL(k_maxpool3x3s2_bwd, dim3(grid_1d(total)), dim3(256), 0,
+                     (hipStream_t)stream, (const bf16_t*)dy,
+                     (const unsigned char*)argmax, (bf16_t*)dx, N, H, W, C, Ho, Wo);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_gap_fwd(const void* x, void* y, int N, int HW, int C,
+                             void* stream) {
+  hipLaunchKernelGGL(k_gap_fwd, dim3(N), dim3(256), 0, (hipStream_t)stream,
+                     (const bf16_t*)x, (bf16_t*)y, HW, C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_gap_bwd(const void* dy, void* dx, int N, int HW, int C,
+                             void* stream) {
+  long total = (long)N * HW * (C >> 3);
+  hipLaunchKernelGGL(k_gap_bwd, dim3(grid_1d(total)), dim3(256), 0,
+                     (hipStream_t)stream, (const bf16_t*)dy, (bf16_t*)dx, HW, C,
+                     total);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_softmax_ce(const void* logits, const void* labels,
+                                void* dlogits, void* loss_sum, int B, int K,
+                                float grad_scale, void* stream) {
+  int waves_needed = B;
+  int blocks = min((waves_needed + 3) / 4, 1024);
+  hipLaunchKernelGGL(k_softmax_ce, dim3(blocks), dim3(256), 0, (hipStream_t)stream,
+                     (const float*)logits, (const long*)labels, (float*)dlogits,
+                     (float*)loss_sum, B, K, grad_scale);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_fused_sgd(const void* chunks, int nchunks, long max_numel,
+                               float lr, float momentum, float weight_decay,
+                               int first_step, void* stream) {
+  dim3 grid(grid_1d(max_numel, 256, 512), min(nchunks, 64));
+  hipLaunchKernelGGL(k_fused_sgd, grid, dim3(256), 0, (hipStream_t)stream,
+                     (const SgdChunk*)chunks, nchunks, lr, momentum,
+                     weight_decay, first_step);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_normalize_u8(const void* x, void* y, long n, void* stream) {
+  hipLaunchKernelGGL(k_normalize_u8, dim3(grid_1d(n / 16)), dim3(256), 0,
+                     (hipStream_t)stream, (const unsigned char*)x, (bf16_t*)y,
+                     n / 16);
+  DDLW_CHECK_LAUNCH();
+}

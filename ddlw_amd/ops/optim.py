@@ -1,0 +1,68 @@
+"""Fused SGD optimizer — one multi-tensor kernel per step (north-star
+"fused SGD step"; replaces torch.optim.SGD's per-tensor foreach ops).
+
+Momentum + weight decay, fp32 params/grads/momentum (momentum buffers start
+at zero, so v = mu*v + g reproduces torch.optim.SGD's buf-initialized-to-grad
+first step). The chunk descriptor table (param/grad/momentum pointers +
+sizes) is built once per group and refreshed only if grad storage moves, so
+the steady-state step is one kernel launch per group.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Tuple
+
+import torch
+
+from . import binding
+
+
+class FusedSGD(torch.optim.Optimizer):
+    def __init__(self, params, lr: float = 0.1, momentum: float = 0.9,
+                 weight_decay: float = 0.0):
+        defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self._desc: Dict[int, Tuple[tuple, torch.Tensor, int]] = {}
+
+    def _group_desc(self, gi: int, params: List[torch.Tensor]):
+        key = tuple((p.data_ptr(), p.grad.data_ptr(), p.numel()) for p in params)
+        cached = self._desc.get(gi)
+        if cached is not None and cached[0] == key:
+            return cached[1], cached[2]
+        rows = []
+        max_numel = 0
+        for p in params:
+            st = self.state[p]
+            if "momentum_buffer" not in st:
+                st["momentum_buffer"] = torch.zeros_like(p)
+            rows.append(
+                (p.data_ptr(), p.grad.data_ptr(), st["momentum_buffer"].data_ptr(),
+                 0, p.numel())
+            )
+            max_numel = max(max_numel, p.numel())
+        desc = torch.tensor(rows, dtype=torch.int64).to(params[0].device)
+        self._desc[gi] = (key, desc, max_numel)
+        return desc, max_numel
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        for gi, group in enumerate(self.param_groups):
+            params = [p for p in group["params"] if p.grad is not None]
+            if not params:
+                continue
+            if not params[0].is_cuda:
+                # CPU fallback = plain SGD-with-momentum semantics (oracle)
+                for p in params:
+                    st = self.state[p]
+                    if "momentum_buffer" not in st:
+                        st["momentum_buffer"] = torch.zeros_like(p)
+                    g = p.grad + group["weight_decay"] * p
+                    st["momentum_buffer"].mul_(group["momentum"]).add_(g)
+                    p.add_(st["momentum_buffer"], alpha=-group["lr"])
+                continue
+            desc, max_numel = self._group_desc(gi, params)
+            binding.fused_sgd(
+                desc, len(params), max_numel, group["lr"], group["momentum"],
+                group["weight_decay"], False,
+            )
+        return loss

@@ -1,0 +1,229 @@
+"""GPU kernel parity tests: each hand-written HIP op vs a plain PyTorch fp32
+reference of the same op. All marked @pytest.mark.gpu.
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+def _cuda():
+    return torch.device("cuda:0")
+
+
+def _cl(t):
+    return t.contiguous(memory_format=torch.channels_last)
+
+
+def _nhwc_bf16(*shape, seed=0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    t = torch.randn(*shape, generator=g).to(_cuda()).to(torch.bfloat16)
+    return _cl(t)
+
+
+BN_SHAPES = [(4, 64, 56, 56), (2, 256, 28, 28), (3, 2048, 7, 7), (2, 24, 14, 14)]
+
+
+@pytest.mark.parametrize("shape", BN_SHAPES)
+@pytest.mark.parametrize("relu", [False, True])
+def test_bn_act_forward_backward_parity(shape, relu):
+    from ddlw_amd.ops.layers import BatchNormAct2d
+
+    n, c, h, w = shape
+    torch.manual_seed(0)
+    x = _nhwc_bf16(n, c, h, w, seed=1)
+    x32 = x.float().detach().requires_grad_(True)
+    xb = x.detach().requires_grad_(True)
+
+    layer = BatchNormAct2d(c, relu=relu).to(_cuda())
+    layer.train()
+    with torch.no_grad():
+        layer.weight.uniform_(0.5, 1.5)
+        layer.bias.uniform_(-0.5, 0.5)
+    ref = torch.nn.BatchNorm2d(c).to(_cuda())
+    ref.load_state_dict({k: v for k, v in layer.state_dict().items()}, strict=False)
+    ref.train()
+
+    y = layer(xb)
+    y_ref = ref(x32)
+    if relu:
+        y_ref = F.relu(y_ref)
+    assert torch.allclose(y.float(), y_ref, atol=5e-2, rtol=5e-2)
+
+    dy = torch.randn_like(y_ref)
+    y.backward(dy.to(torch.bfloat16))
+    y_ref.backward(dy)
+    assert torch.allclose(xb.grad.float(), x32.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(layer.weight.grad, ref.weight.grad, atol=2e-1, rtol=2e-2)
+    assert torch.allclose(layer.bias.grad, ref.bias.grad, atol=2e-1, rtol=2e-2)
+    # running stats updated like the stock op
+    assert torch.allclose(layer.running_mean, ref.running_mean, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(layer.running_var, ref.running_var, atol=1e-2, rtol=1e-2)
+
+
+def test_bn_residual_add_relu_parity():
+    from ddlw_amd.ops.layers import BatchNormAct2d
+
+    n, c, h, w = 4, 128, 14, 14
+    x = _nhwc_bf16(n, c, h, w, seed=2)
+    r = _nhwc_bf16(n, c, h, w, seed=3)
+    xb = x.detach().requires_grad_(True)
+    rb = r.detach().requires_grad_(True)
+    x32 = x.float().detach().requires_grad_(True)
+    r32 = r.float().detach().requires_grad_(True)
+
+    layer = BatchNormAct2d(c, relu=True).to(_cuda()).train()
+    ref = torch.nn.BatchNorm2d(c).to(_cuda()).train()
+
+    y = layer(xb, residual=rb)
+    y_ref = F.relu(ref(x32) + r32)
+    assert torch.allclose(y.float(), y_ref, atol=5e-2, rtol=5e-2)
+
+    dy = torch.randn_like(y_ref)
+    y.backward(dy.to(torch.bfloat16))
+    y_ref.backward(dy)
+    assert torch.allclose(xb.grad.float(), x32.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(rb.grad.float(), r32.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_bn_eval_mode_parity():
+    from ddlw_amd.ops.layers import BatchNormAct2d
+
+    n, c, h, w = 2, 64, 28, 28
+    layer = BatchNormAct2d(c, relu=True).to(_cuda())
+    with torch.no_grad():
+        layer.running_mean.uniform_(-0.3, 0.3)
+        layer.running_var.uniform_(0.7, 1.4)
+    ref = torch.nn.BatchNorm2d(c).to(_cuda())
+    ref.load_state_dict(dict(layer.state_dict()), strict=False)
+    layer.eval(), ref.eval()
+    x = _nhwc_bf16(n, c, h, w, seed=5)
+    xb = x.detach().requires_grad_(True)
+    x32 = x.float().detach().requires_grad_(True)
+    y = layer(xb)
+    y_ref = F.relu(ref(x32))
+    assert torch.allclose(y.float(), y_ref, atol=5e-2, rtol=5e-2)
+    dy = torch.randn_like(y_ref)
+    y.backward(dy.to(torch.bfloat16))
+    y_ref.backward(dy)
+    assert torch.allclose(xb.grad.float(), x32.grad, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("shape", [(2, 64, 112, 112), (3, 64, 57, 57)])
+def test_maxpool_parity(shape):
+    from ddlw_amd.ops.layers import MaxPool3x3s2
+
+    n, c, h, w = shape
+    x = _nhwc_bf16(n, c, h, w, seed=7)
+    xb = x.detach().requires_grad_(True)
+    x32 = x.float().detach().requires_grad_(True)
+    y = MaxPool3x3s2()(xb)
+    y_ref = F.max_pool2d(x32, 3, stride=2, padding=1)
+    assert y.shape == y_ref.shape
+    assert torch.allclose(y.float(), y_ref, atol=1e-2, rtol=1e-2)
+    dy = torch.randn_like(y_ref)
+    y.backward(dy.to(torch.bfloat16))
+    y_ref.backward(dy)
+    assert torch.allclose(xb.grad.float(), x32.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_gap_parity():
+    from ddlw_amd.ops.layers import GlobalAvgPool2d
+
+    x = _nhwc_bf16(4, 2048, 7, 7, seed=9)
+    xb = x.detach().requires_grad_(True)
+    x32 = x.float().detach().requires_grad_(True)
+    y = GlobalAvgPool2d()(xb)
+    y_ref = F.adaptive_avg_pool2d(x32, 1).flatten(1)
+    assert torch.allclose(y.float(), y_ref, atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(y_ref)
+    y.backward(dy.to(torch.bfloat16))
+    y_ref.backward(dy)
+    assert torch.allclose(xb.grad.float(), x32.grad, atol=2e-2, rtol=2e-2)
+
+
+def test_softmax_ce_parity():
+    from ddlw_amd.ops.layers import softmax_cross_entropy
+
+    torch.manual_seed(11)
+    logits = torch.randn(64, 1000, device=_cuda()) * 4
+    labels = torch.randint(0, 1000, (64,), device=_cuda())
+    l32 = logits.detach().requires_grad_(True)
+    lb = logits.detach().requires_grad_(True)
+    loss = softmax_cross_entropy(lb, labels)
+    loss_ref = F.cross_entropy(l32, labels)
+    assert torch.allclose(loss, loss_ref, atol=1e-4, rtol=1e-4)
+    loss.backward()
+    loss_ref.backward()
+    assert torch.allclose(lb.grad, l32.grad, atol=1e-5, rtol=1e-4)
+
+
+def test_fused_sgd_matches_torch_sgd():
+    from ddlw_amd.ops.optim import FusedSGD
+
+    torch.manual_seed(13)
+    p1 = [torch.randn(1000, device=_cuda(), requires_grad=True) for _ in range(3)]
+    p2 = [p.detach().clone().requires_grad_(True) for p in p1]
+    o1 = FusedSGD(p1, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    o2 = torch.optim.SGD(p2, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    for step in range(3):
+        g = [torch.randn(1000, device=_cuda()) for _ in range(3)]
+        for a, b, gg in zip(p1, p2, g):
+            a.grad = gg.clone()
+            b.grad = gg.clone()
+        o1.step()
+        o2.step()
+    for a, b in zip(p1, p2):
+        assert torch.allclose(a, b, atol=1e-5, rtol=1e-5)
+
+
+def test_normalize_u8_parity():
+    from ddlw_amd.ops.layers import normalize_u8_bf16
+
+    x = torch.randint(0, 256, (2, 3, 32, 32), dtype=torch.uint8).contiguous(
+        memory_format=torch.channels_last
+    ).to(_cuda())
+    y = normalize_u8_bf16(x)
+    ref = (x.float() / 127.5) - 1.0
+    assert y.dtype == torch.bfloat16
+    assert torch.allclose(y.float(), ref, atol=1e-2, rtol=1e-2)
+
+
+def test_resnet50_hip_step_runs_and_matches_loss_scale():
+    """End-to-end: one fwd+bwd+opt step of ResNet-50 with every HIP op active;
+    loss must be finite and near ln(1000) at random init."""
+    import math
+
+    from ddlw_amd.models import build_resnet50
+    from ddlw_amd.ops import FusedSGD, softmax_cross_entropy
+
+    torch.manual_seed(17)
+    model = build_resnet50(num_classes=1000).to(_cuda()).to(memory_format=torch.channels_last)
+    opt = FusedSGD(model.parameters(), lr=0.01, momentum=0.9)
+    x = _nhwc_bf16(8, 3, 224, 224, seed=19)
+    ylab = torch.randint(0, 1000, (8,), device=_cuda())
+    model.train()
+    losses = []
+    for _ in range(3):
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            logits = model(x)
+        loss = softmax_cross_entropy(logits, ylab)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert all(math.isfinite(l) for l in losses)
+    assert abs(losses[0] - math.log(1000)) < 1.0
+    assert losses[-1] < losses[0] + 0.5  # training on a fixed batch shouldn't blow up
+
+
+def test_native_lib_is_loaded():
+    """The round-end check looks for the in-tree .so in loaded maps — assert
+    the HIP library really is what runs (no silent eager fallback)."""
+    from ddlw_amd.ops import require_lib
+    from ddlw_amd.ops.runtime import LIB_DIR, LIB_NAME
+
+    require_lib()
+    maps = open("/proc/self/maps").read()
+    assert str(LIB_DIR / LIB_NAME) in maps
