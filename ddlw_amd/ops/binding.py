@@ -40,20 +40,25 @@ def supported_channels(c: int) -> bool:
     return c % 8 == 0
 
 
+def _nparts(rows: int, C: int) -> int:
+    lib = require_lib()
+    return int(lib.ddlw_bn_nparts(ctypes.c_long(rows), ctypes.c_int(C)))
+
+
 def bn_stats(x: torch.Tensor, eps: float, momentum: float,
              running_mean: Optional[torch.Tensor], running_var: Optional[torch.Tensor]):
     lib = require_lib()
     rows, C = _rows_c(x)
     dev = x.device
-    sum_ = torch.zeros(C, dtype=torch.float32, device=dev)
-    sumsq = torch.zeros(C, dtype=torch.float32, device=dev)
+    ny = _nparts(rows, C)
+    part = torch.empty(2, ny, C, dtype=torch.float32, device=dev)
     mean = torch.empty(C, dtype=torch.float32, device=dev)
     rstd = torch.empty(C, dtype=torch.float32, device=dev)
     s = current_stream_ptr()
-    check(lib.ddlw_bn_stats(_nhwc_ptr(x), _p(sum_), _p(sumsq),
+    check(lib.ddlw_bn_stats(_nhwc_ptr(x), _p(part[0]), _p(part[1]),
                             ctypes.c_long(rows), ctypes.c_int(C),
                             ctypes.c_void_p(s)), "bn_stats")
-    check(lib.ddlw_bn_finalize(_p(sum_), _p(sumsq), _p(mean), _p(rstd),
+    check(lib.ddlw_bn_finalize(_p(part[0]), _p(part[1]), _p(mean), _p(rstd),
                                _p(running_mean), _p(running_var),
                                ctypes.c_long(rows), ctypes.c_int(C),
                                ctypes.c_float(eps), ctypes.c_float(momentum),
@@ -81,15 +86,25 @@ def bn_bwd_reduce(dy, y, x, mean, rstd, relu: bool):
     lib = require_lib()
     rows, C = _rows_c(x)
     dev = x.device
-    dbeta = torch.zeros(C, dtype=torch.float32, device=dev)
-    dgamma = torch.zeros(C, dtype=torch.float32, device=dev)
+    ny = _nparts(rows, C)
+    part = torch.empty(2, ny, C, dtype=torch.float32, device=dev)
+    dbeta = torch.empty(C, dtype=torch.float32, device=dev)
+    dgamma = torch.empty(C, dtype=torch.float32, device=dev)
+    s = current_stream_ptr()
     check(
         lib.ddlw_bn_bwd_reduce(
-            _nhwc_ptr(dy), _p(y), _nhwc_ptr(x), _p(mean), _p(rstd), _p(dbeta),
-            _p(dgamma), ctypes.c_long(rows), ctypes.c_int(C),
-            ctypes.c_int(1 if relu else 0), ctypes.c_void_p(current_stream_ptr())
+            _nhwc_ptr(dy), _p(y), _nhwc_ptr(x), _p(mean), _p(rstd), _p(part[0]),
+            _p(part[1]), ctypes.c_long(rows), ctypes.c_int(C),
+            ctypes.c_int(1 if relu else 0), ctypes.c_void_p(s)
         ),
         "bn_bwd_reduce",
+    )
+    check(
+        lib.ddlw_bn_grad_finalize(
+            _p(part[0]), _p(part[1]), _p(dbeta), _p(dgamma), ctypes.c_long(rows),
+            ctypes.c_int(C), ctypes.c_void_p(s)
+        ),
+        "bn_grad_finalize",
     )
     return dbeta, dgamma
 

@@ -31,12 +31,16 @@ class Conv2d(nn.Conv2d):
             and x.is_cuda
             and x.dtype == torch.bfloat16
             and self.groups == 1
+            and self.dilation == (1, 1)
+            and os.environ.get("DDLW_DISABLE_HIP_OPS", "0") != "1"
         ):
             from . import conv_gemm
 
             if conv_gemm.available(self, x, mode):
+                hip_fwd, hip_dgrad = self._ddlw_route
                 return conv_gemm.conv2d(
-                    x, self.weight, self.bias, self.stride, self.padding
+                    x, self.weight, self.bias, self.stride, self.padding,
+                    hip_fwd, hip_dgrad,
                 )
         return F.conv2d(
             x, self.weight.to(x.dtype),

@@ -23,16 +23,18 @@ DDLW_EXPORT int ddlw_device_sync() {
 
 // ---------------------------------------------------------------------------
 // BatchNorm training statistics: per-channel sum / sumsq over rows = N*H*W.
-// x viewed as [rows][C]; C % 8 == 0 and C/8 is a power of two <= 256
-// (every ResNet-50 / MobileNetV2 channel count satisfies this; the generic
-// fallback tile loop covers other C).
-// Each thread owns 8 consecutive channels (one bf16x8 vector); thread groups
-// stack over rows; LDS tree-reduce across row groups; one fp32 atomicAdd per
-// channel into the zero-initialised global accumulators.
+// x viewed as [rows][C]; C % 8 == 0. Each thread owns 8 consecutive channels
+// (one bf16x8 vector); thread groups stack over rows; LDS tree-reduce across
+// row groups; each block writes its PARTIAL sums to part[blockIdx.y][C]
+// (no atomics — fp32 atomicAdd contention on the small per-channel arrays
+// measured 3.5x off the HBM roofline — and fully deterministic); the
+// finalize kernel reduces the <=BN_MAX_PARTS partials.
 // ---------------------------------------------------------------------------
+#define BN_MAX_PARTS 512  // 2 blocks/CU of 256 threads = 8 waves/CU in flight
+
 __global__ __launch_bounds__(256) void k_bn_stats(
-    const bf16_t* __restrict__ x, float* __restrict__ sum, float* __restrict__ sumsq,
-    long rows, int C) {
+    const bf16_t* __restrict__ x, float* __restrict__ part_sum,
+    float* __restrict__ part_sumsq, long rows, int C) {
   const int vecC = C >> 3;
   const int VPB = vecC < 256 ? vecC : 256;      // vectors per block
   const int ROWG = 256 / VPB;                   // row groups per block (floor)
@@ -78,24 +80,31 @@ __global__ __launch_bounds__(256) void k_bn_stats(
     }
   }
   if (active && rowg == 0) {
+    float* ps = part_sum + (long)blockIdx.y * C;
+    float* pq = part_sumsq + (long)blockIdx.y * C;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      atomicAdd(sum + vec * 8 + k, s[k]);
-      atomicAdd(sumsq + vec * 8 + k, q[k]);
+      ps[vec * 8 + k] = s[k];
+      pq[vec * 8 + k] = q[k];
     }
   }
 }
 
-// finalize: mean/rstd + running-stat update (train-mode BN, K4 fwd+bwd row)
+// finalize: reduce partials -> mean/rstd + running-stat update (K4)
 __global__ __launch_bounds__(256) void k_bn_finalize(
-    const float* __restrict__ sum, const float* __restrict__ sumsq,
+    const float* __restrict__ part_sum, const float* __restrict__ part_sumsq,
     float* __restrict__ mean, float* __restrict__ rstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
-    long rows, int C, float eps, float momentum) {
+    long rows, int C, int nparts, float eps, float momentum) {
   int c = blockIdx.x * 256 + threadIdx.x;
   if (c >= C) return;
-  float m = sum[c] / (float)rows;
-  float var = fmaxf(sumsq[c] / (float)rows - m * m, 0.f);
+  float s = 0.f, q = 0.f;
+  for (int p = 0; p < nparts; ++p) {
+    s += part_sum[(long)p * C + c];
+    q += part_sumsq[(long)p * C + c];
+  }
+  float m = s / (float)rows;
+  float var = fmaxf(q / (float)rows - m * m, 0.f);
   mean[c] = m;
   rstd[c] = rsqrtf(var + eps);
   if (running_mean) {
@@ -199,12 +208,29 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
         for (int k = 0; k < 8; ++k) dg[k] += lds[(g * VPB + (tid % VPB)) * 8 + k];
   }
   if (active && rowg == 0) {
+    float* pb = dbeta + (long)blockIdx.y * C;   // partial rows, finalized below
+    float* pg = dgamma + (long)blockIdx.y * C;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      atomicAdd(dbeta + vec * 8 + k, db[k]);
-      atomicAdd(dgamma + vec * 8 + k, dg[k]);
+      pb[vec * 8 + k] = db[k];
+      pg[vec * 8 + k] = dg[k];
     }
   }
+}
+
+// reduce the bwd partials -> dbeta[C], dgamma[C]
+__global__ __launch_bounds__(256) void k_bn_grad_finalize(
+    const float* __restrict__ part_db, const float* __restrict__ part_dg,
+    float* __restrict__ dbeta, float* __restrict__ dgamma, int C, int nparts) {
+  int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= C) return;
+  float b = 0.f, g = 0.f;
+  for (int p = 0; p < nparts; ++p) {
+    b += part_db[(long)p * C + c];
+    g += part_dg[(long)p * C + c];
+  }
+  dbeta[c] = b;
+  dgamma[c] = g;
 }
 
 // ---------------------------------------------------------------------------
@@ -479,24 +505,37 @@ static inline int grid_1d(long work, int block = 256, int cap = 2048) {
   return (int)(g < cap ? (g > 0 ? g : 1) : cap);
 }
 
-DDLW_EXPORT int ddlw_bn_stats(const void* x, void* sum, void* sumsq,
-                              long rows, int C, void* stream) {
+// shared geometry for the BN-style row reductions; ny (= partial count) must
+// match between the reduce launch and its finalize
+DDLW_EXPORT int ddlw_bn_nparts(long rows, int C) {
   int vecC = C >> 3;
   int VPB = vecC < 256 ? vecC : 256;
   int ROWG = 256 / VPB;
-  dim3 grid((vecC + VPB - 1) / VPB, (int)min((rows + ROWG - 1) / ROWG, (long)1024));
+  long ny = (rows + ROWG - 1) / ROWG;
+  if (ny > BN_MAX_PARTS) ny = BN_MAX_PARTS;
+  return (int)ny;
+}
+
+DDLW_EXPORT int ddlw_bn_stats(const void* x, void* part_sum, void* part_sumsq,
+                              long rows, int C, void* stream) {
+  int vecC = C >> 3;
+  int VPB = vecC < 256 ? vecC : 256;
+  dim3 grid((vecC + VPB - 1) / VPB, ddlw_bn_nparts(rows, C));
   hipLaunchKernelGGL(k_bn_stats, grid, dim3(256), 0, (hipStream_t)stream,
-                     (const bf16_t*)x, (float*)sum, (float*)sumsq, rows, C);
+                     (const bf16_t*)x, (float*)part_sum, (float*)part_sumsq,
+                     rows, C);
   DDLW_CHECK_LAUNCH();
 }
 
-DDLW_EXPORT int ddlw_bn_finalize(const void* sum, const void* sumsq, void* mean,
-                                 void* rstd, void* rmean, void* rvar, long rows,
-                                 int C, float eps, float momentum, void* stream) {
+DDLW_EXPORT int ddlw_bn_finalize(const void* part_sum, const void* part_sumsq,
+                                 void* mean, void* rstd, void* rmean, void* rvar,
+                                 long rows, int C, float eps, float momentum,
+                                 void* stream) {
   hipLaunchKernelGGL(k_bn_finalize, dim3((C + 255) / 256), dim3(256), 0,
-                     (hipStream_t)stream, (const float*)sum, (const float*)sumsq,
-                     (float*)mean, (float*)rstd, (float*)rmean, (float*)rvar,
-                     rows, C, eps, momentum);
+                     (hipStream_t)stream, (const float*)part_sum,
+                     (const float*)part_sumsq, (float*)mean, (float*)rstd,
+                     (float*)rmean, (float*)rvar, rows, C,
+                     ddlw_bn_nparts(rows, C), eps, momentum);
   DDLW_CHECK_LAUNCH();
 }
 
@@ -535,8 +574,7 @@ DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* y, const void* x,
                                    int relu, void* stream) {
   int vecC = C >> 3;
   int VPB = vecC < 256 ? vecC : 256;
-  int ROWG = 256 / VPB;
-  dim3 grid((vecC + VPB - 1) / VPB, (int)min((rows + ROWG - 1) / ROWG, (long)1024));
+  dim3 grid((vecC + VPB - 1) / VPB, ddlw_bn_nparts(rows, C));
   if (relu)
     hipLaunchKernelGGL((k_bn_bwd_reduce<1>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
@@ -547,6 +585,16 @@ DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* y, const void* x,
                        (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (float*)dbeta,
                        (float*)dgamma, rows, C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_bn_grad_finalize(const void* part_db, const void* part_dg,
+                                      void* dbeta, void* dgamma, long rows,
+                                      int C, void* stream) {
+  hipLaunchKernelGGL(k_bn_grad_finalize, dim3((C + 255) / 256), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)part_db,
+                     (const float*)part_dg, (float*)dbeta, (float*)dgamma, C,
+                     ddlw_bn_nparts(rows, C));
   DDLW_CHECK_LAUNCH();
 }
 
