@@ -4,89 +4,129 @@
 // fwd/bwd ... MFMA implicit-GEMM with LDS im2col tiles"):
 //   O[M=N*Ho*Wo][Cout] = im2col(x)[M][R*S*C] @ W[Cout][R*S*C]^T
 // Weight layout = channels_last flat [Cout][R][S][C] (B^T GEMM input).
-// Loop order r,s outer / C-chunks inner, so every staged 8-vector is 8
+// Loop order r,s outer / C-chunks inner, so every staged 16-B chunk is 8
 // consecutive channels of one pixel (no divisions in the hot loop).
 //
 // dgrad(stride=1) reuses THIS kernel with flipped/transposed weights
 // (dx = conv_s1(dy, W'); W'[c][r'][s'][k] = W[k][R-1-r'][S-1-s'][c],
 // pad' = R-1-pad), so fwd and dgrad share one MFMA path.
 //
-// Tiles: BM x BN x BK=32, 4 waves (256 threads) in a 2x2 wave grid, each
-// wave a (BM/2 x BN/2) sub-tile of 16x16 fragments via
-// v_mfma_f32_16x16x32_bf16 (one MFMA covers the whole BK=32 K-step).
-// LDS rows padded to 40 bf16 (80 B) -> conflict-free ds_read_b128 column
-// reads (16 lanes x stride-20-dword rows cover 16 distinct banks).
+// Structure = the cdna_hip_programming.md §5 "step-3" ladder shape:
+//   - BK=64 K-steps, TWO LDS buffers, async global->LDS via
+//     global_load_lds_dwordx4 (1 KiB per wave-instruction), plain
+//     __syncthreads() per step (hipcc adds the vmcnt drain);
+//   - LDS image is lane-linear [rows][64] bf16 (glds writes base+lane*16),
+//     so the bank swizzle lives on the SOURCE address and the fragment
+//     read: chunk col8 ^= (row & 7)  (rule 21: same involution both sides);
+//   - im2col padding/tile edges are handled by redirecting the per-lane
+//     glds source to a zero page (glds has no predication);
+//   - 2x2 wave grid, each wave (BM/2 x BN/2) of 16x16 fragments,
+//     v_mfma_f32_16x16x32_bf16, two K-halves per BK=64 step;
+//   - XCD-aware bijective workgroup swizzle (L2 tile locality).
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4_v;
 
-#define LDSP 40  // padded row stride in bf16 elements (32 + 8)
+#define GLOBAL_AS __attribute__((address_space(1)))
+#define LDS_AS __attribute__((address_space(3)))
 
 template <int BM, int BN>
-__global__ __launch_bounds__(256, 2) void k_conv_fwd_igemm(
+__global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
-    bf16_t* __restrict__ y,
+    bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
     int N, int H, int W_, int C, int K, int Ho, int Wo,
-    int R, int S, int stride, int pad) {
-  constexpr int WAVES_M = 2, WAVES_N = 2;
-  constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;  // per-wave tile
-  constexpr int MF = WM / 16, NF = WN / 16;            // fragments per wave
-  constexpr int BK = 32;
+    int R, int S, int stride, int pad, int nwg_swz) {
+  constexpr int BK = 64;
+  constexpr int WM = BM / 2, WN = BN / 2;
+  constexpr int MF = WM / 16, NF = WN / 16;
+  constexpr int AP = BM / 32;  // 1-KiB A pieces per wave (8 rows each)
+  constexpr int BP = BN / 32;  // 1-KiB B pieces per wave
+  constexpr int BUF = (BM + BN) * BK;  // bf16 elements per buffer
 
-  __shared__ __attribute__((aligned(16))) bf16_t lds[(BM + BN) * LDSP];
-  bf16_t* lA = lds;              // [BM][LDSP]
-  bf16_t* lB = lds + BM * LDSP;  // [BN][LDSP]
+  __shared__ __attribute__((aligned(16))) bf16_t smem[2 * BUF];
 
   const long M = (long)N * Ho * Wo;
   const int tiles_n = (K + BN - 1) / BN;
-  // grid.x enumerates (tile_m, tile_n); XCD-aware swizzle happens host-side
-  const int tile_n = blockIdx.x % tiles_n;
-  const long tile_m = blockIdx.x / tiles_n;
+
+  // bijective XCD swizzle: consecutive swizzled ids land on one XCD
+  int wg = blockIdx.x;
+  {
+    int nwg = nwg_swz;
+    int q = nwg >> 3, rm = nwg & 7;
+    int xcd = wg & 7, i = wg >> 3;
+    wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + i;
+  }
+  const int tile_n = wg % tiles_n;
+  const long tile_m = wg / tiles_n;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wr = wave >> 1, wc = wave & 1;
 
-  // ---- staging geometry: thread t stages vectors {t, t+256, ...} of the
-  // A tile (BM rows x 4 col8) and B tile (BN rows x 4 col8)
-  constexpr int AVEC = BM * 4, BVEC = BN * 4;
-  constexpr int AIT = (AVEC + 255) / 256, BIT = (BVEC + 255) / 256;
-  // per-thread A rows are fixed: v/4 for v in {tid, tid+256}
-  int a_row[AIT], a_col8[AIT];
-  long a_m[AIT];
-  int a_hb[AIT], a_wb[AIT];
-  const bf16_t* a_base[AIT];
+  // ---- per-thread staging geometry (fixed across k-steps) -----------------
+  // piece pa = wave*AP + p covers LDS rows pa*8..pa*8+7; this lane's row/col:
+  const int prow = lane >> 3;          // row within piece
+  const int pcol8 = lane & 7;          // dest col8 within row
+  long a_m[AP];
+  int a_hb[AP], a_wb[AP];
+  const bf16_t* a_base[AP];
   #pragma unroll
-  for (int i = 0; i < AIT; ++i) {
-    int v = tid + i * 256;
-    a_row[i] = v >> 2;
-    a_col8[i] = v & 3;
-    long m = tile_m * BM + a_row[i];
-    a_m[i] = m;
+  for (int p = 0; p < AP; ++p) {
+    int row = (wave * AP + p) * 8 + prow;
+    long m = tile_m * BM + row;
+    a_m[p] = m;
     if (m < M) {
       int wo = (int)(m % Wo);
       long t2 = m / Wo;
       int ho = (int)(t2 % Ho);
       int n = (int)(t2 / Ho);
-      a_hb[i] = ho * stride - pad;
-      a_wb[i] = wo * stride - pad;
-      a_base[i] = x + (((long)n * H + a_hb[i]) * W_ + a_wb[i]) * C;
+      a_hb[p] = ho * stride - pad;
+      a_wb[p] = wo * stride - pad;
+      // source chunk col8' = dest col8 ^ (row&7) = pcol8 ^ prow (swizzle)
+      a_base[p] = x + (((long)n * H + a_hb[p]) * W_ + a_wb[p]) * C +
+                  (pcol8 ^ prow) * 8;
     } else {
-      a_hb[i] = -100000;  // never valid
-      a_wb[i] = -100000;
-      a_base[i] = x;
+      a_hb[p] = -100000;
+      a_wb[p] = -100000;
+      a_base[p] = zpage;
     }
   }
-  int b_row[BIT], b_col8[BIT];
-  #pragma unroll
-  for (int i = 0; i < BIT; ++i) {
-    int v = tid + i * 256;
-    b_row[i] = (v < BVEC) ? (v >> 2) : -1;
-    b_col8[i] = v & 3;
-  }
   const long KRS = (long)R * S * C;
+  const bf16_t* b_base[BP];
+  bool b_ok[BP];
+  #pragma unroll
+  for (int p = 0; p < BP; ++p) {
+    int row = (wave * BP + p) * 8 + prow;
+    int j = tile_n * BN + row;
+    b_ok[p] = j < K;
+    b_base[p] = b_ok[p] ? (w + (long)j * KRS + (pcol8 ^ prow) * 8) : zpage;
+  }
+
+  // ---- staging: one glds per piece into buffer `b` for k-step (r,s,ck)
+  auto stage = [&](int buf, int r, int s, int ck) {
+    bf16_t* lA = smem + buf * BUF;
+    bf16_t* lB = lA + BM * BK;
+    const long aoff = ((long)r * W_ + s) * C + ck;
+    #pragma unroll
+    for (int p = 0; p < AP; ++p) {
+      int h = a_hb[p] + r, ww = a_wb[p] + s;
+      bool ok = (a_m[p] < M) & (h >= 0) & (h < H) & (ww >= 0) & (ww < W_);
+      const bf16_t* src = ok ? (a_base[p] + aoff) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src,
+          (LDS_AS void*)(lA + (wave * AP + p) * 512), 16, 0, 0);
+    }
+    const long boff = ((long)r * S + s) * C + ck;
+    #pragma unroll
+    for (int p = 0; p < BP; ++p) {
+      const bf16_t* src = b_ok[p] ? (b_base[p] + boff) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src,
+          (LDS_AS void*)(lB + (wave * BP + p) * 512), 16, 0, 0);
+    }
+  };
 
   f32x4_v acc[MF][NF];
   #pragma unroll
@@ -94,61 +134,59 @@ __global__ __launch_bounds__(256, 2) void k_conv_fwd_igemm(
     #pragma unroll
     for (int ni = 0; ni < NF; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
-  // fragment read offsets (bf16 elements, padded rows)
-  const int fr_row = lane & 15;        // row within 16-frag
-  const int fr_k8 = (lane >> 4) * 8;   // k offset (8 bf16 per lane)
+  const int fr_row = lane & 15;
+  const int fr_c8 = lane >> 4;  // col8 base per k-half: kh*4 + (lane>>4)
 
-  for (int r = 0; r < R; ++r) {
-    for (int s = 0; s < S; ++s) {
-      for (int ck = 0; ck < C; ck += BK) {
-        // ---- stage A (predicated; zeros for pad/out-of-range)
-        #pragma unroll
-        for (int i = 0; i < AIT; ++i) {
-          int h = a_hb[i] + r, ww = a_wb[i] + s;
-          int c = ck + a_col8[i] * 8;
-          uint4 val = {0, 0, 0, 0};
-          if (a_m[i] < M && h >= 0 && h < H && ww >= 0 && ww < W_) {
-            val = *reinterpret_cast<const uint4*>(
-                a_base[i] + ((long)r * W_ + s) * C + c);
-          }
-          *reinterpret_cast<uint4*>(lA + a_row[i] * LDSP + a_col8[i] * 8) = val;
-        }
-        // ---- stage B
-        #pragma unroll
-        for (int i = 0; i < BIT; ++i) {
-          if (b_row[i] < 0) continue;
-          int j = tile_n * BN + b_row[i];
-          int c = ck + b_col8[i] * 8;
-          uint4 val = {0, 0, 0, 0};
-          if (j < K) {
-            val = *reinterpret_cast<const uint4*>(
-                w + (long)j * KRS + ((long)r * S + s) * C + c);
-          }
-          *reinterpret_cast<uint4*>(lB + b_row[i] * LDSP + b_col8[i] * 8) = val;
-        }
-        __syncthreads();
-        // ---- fragments + MFMA
-        bf16x8_v fa[MF], fb[NF];
-        #pragma unroll
-        for (int mi = 0; mi < MF; ++mi)
-          fa[mi] = *reinterpret_cast<const bf16x8_v*>(
-              lA + (wr * WM + mi * 16 + fr_row) * LDSP + fr_k8);
+  const int csteps = C / BK;
+  const int T = R * S * csteps;
+  int r2 = 0, s2 = 0, ck2 = 0;
+  auto advance = [&]() {
+    ck2 += BK;
+    if (ck2 >= C) {
+      ck2 = 0;
+      if (++s2 >= S) { s2 = 0; ++r2; }
+    }
+  };
+
+  stage(0, r2, s2, ck2);
+  advance();
+  __syncthreads();
+
+  int cur = 0;
+  for (int t = 0; t < T; ++t) {
+    if (t + 1 < T) {
+      stage(cur ^ 1, r2, s2, ck2);
+      advance();
+    }
+    bf16_t* lA = smem + cur * BUF;
+    bf16_t* lB = lA + BM * BK;
+    #pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      bf16x8_v fa[MF], fb[NF];
+      #pragma unroll
+      for (int mi = 0; mi < MF; ++mi) {
+        int row = wr * WM + mi * 16 + fr_row;
+        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+        fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
+      }
+      #pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        int row = wc * WN + ni * 16 + fr_row;
+        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+        fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
+      }
+      #pragma unroll
+      for (int mi = 0; mi < MF; ++mi)
         #pragma unroll
         for (int ni = 0; ni < NF; ++ni)
-          fb[ni] = *reinterpret_cast<const bf16x8_v*>(
-              lB + (wc * WN + ni * 16 + fr_row) * LDSP + fr_k8);
-        #pragma unroll
-        for (int mi = 0; mi < MF; ++mi)
-          #pragma unroll
-          for (int ni = 0; ni < NF; ++ni)
-            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
-        __syncthreads();
-      }
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
     }
+    __syncthreads();
+    cur ^= 1;
   }
 
-  // ---- epilogue: D lane map (16x16): col = lane&15, row = (lane>>4)*4 + reg
+  // ---- epilogue: D lane map (16x16): col = lane&15, row = (lane>>4)*4 + q
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
   #pragma unroll
@@ -172,30 +210,27 @@ __global__ __launch_bounds__(256, 2) void k_conv_fwd_igemm(
 static inline long cdiv(long a, long b) { return (a + b - 1) / b; }
 
 DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
+                                    const void* zpage,
                                     int N, int H, int W_, int C, int K,
                                     int Ho, int Wo, int R, int S, int stride,
                                     int pad, void* stream) {
-  if (C % 32 != 0) {
-    ddlw_set_error("conv_fwd_igemm: C must be a multiple of 32");
+  if (C % 64 != 0) {
+    ddlw_set_error("conv_fwd_igemm: C must be a multiple of 64");
     return 2;
   }
   long M = (long)N * Ho * Wo;
   hipStream_t st = (hipStream_t)stream;
-  if (K >= 128) {
-    long grid = cdiv(M, 128) * cdiv(K, 128);
-    hipLaunchKernelGGL((k_conv_fwd_igemm<128, 128>), dim3((int)grid), dim3(256),
-                       0, st, (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y,
-                       N, H, W_, C, K, Ho, Wo, R, S, stride, pad);
-  } else if (K >= 64) {
-    long grid = cdiv(M, 128) * cdiv(K, 64);
-    hipLaunchKernelGGL((k_conv_fwd_igemm<128, 64>), dim3((int)grid), dim3(256),
-                       0, st, (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y,
-                       N, H, W_, C, K, Ho, Wo, R, S, stride, pad);
-  } else {
-    long grid = cdiv(M, 128) * cdiv(K, 32);
-    hipLaunchKernelGGL((k_conv_fwd_igemm<128, 32>), dim3((int)grid), dim3(256),
-                       0, st, (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y,
-                       N, H, W_, C, K, Ho, Wo, R, S, stride, pad);
-  }
+#define LAUNCH(BM, BN)                                                        \
+  do {                                                                        \
+    long grid = cdiv(M, BM) * cdiv(K, BN);                                    \
+    hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN>), dim3((int)grid), dim3(256),\
+                       0, st, (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y, \
+                       (const bf16_t*)zpage, N, H, W_, C, K, Ho, Wo, R, S,    \
+                       stride, pad, (int)grid);                               \
+  } while (0)
+  if (K >= 128) LAUNCH(128, 128);
+  else if (K >= 64) LAUNCH(128, 64);
+  else LAUNCH(128, 32);
+#undef LAUNCH
   DDLW_CHECK_LAUNCH();
 }
