@@ -186,36 +186,19 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     cur ^= 1;
   }
 
-  // ---- epilogue: LDS bounce -> whole-row dwordx4 stores.
-  // D lane map (16x16): col = lane&15, row = (lane>>4)*4 + q. Scalar 2-B
-  // stores straight from that map are store-ISSUE-bound (T21 class); instead
-  // each wave scatters its accs into a private padded LDS tile and stores
-  // whole rows, one row per lane, 16 B per instruction.
+  // ---- epilogue: D lane map (16x16): col = lane&15, row = (lane>>4)*4 + q
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
-  constexpr int WNP = WN + 8;  // pad: row stride off the bank power-of-two
-  bf16_t* lC = smem + wave * (WM * WNP);
   #pragma unroll
-  for (int mi = 0; mi < MF; ++mi)
+  for (int mi = 0; mi < MF; ++mi) {
     #pragma unroll
-    for (int ni = 0; ni < NF; ++ni)
+    for (int ni = 0; ni < NF; ++ni) {
+      int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+      if (j >= K) continue;
       #pragma unroll
-      for (int q = 0; q < 4; ++q)
-        lC[(mi * 16 + d_row0 + q) * WNP + ni * 16 + d_col] =
-            f2b(acc[mi][ni][q]);
-  // wave-private region: lgkmcnt ordering suffices, no barrier needed
-  const long m = tile_m * BM + wr * WM + lane;
-  if (m < M) {
-    const int j_base = tile_n * BN + wc * WN;
-    #pragma unroll
-    for (int v = 0; v < WN / 8; ++v) {
-      int j0 = j_base + v * 8;
-      uint4 val = *reinterpret_cast<const uint4*>(lC + lane * WNP + v * 8);
-      if (j0 + 8 <= K) {
-        *reinterpret_cast<uint4*>(y + m * K + j0) = val;
-      } else {
-        const bf16_t* h = reinterpret_cast<const bf16_t*>(&val);
-        for (int e = 0; e < 8 && j0 + e < K; ++e) y[m * K + j0 + e] = h[e];
+      for (int q = 0; q < 4; ++q) {
+        long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
+        if (m < M) y[m * K + j] = f2b(acc[mi][ni][q]);
       }
     }
   }
