@@ -129,6 +129,8 @@ _active_experiment_id: Optional[str] = None
 def set_experiment(name: str) -> str:
     global _active_experiment_id
     _active_experiment_id = create_experiment(name)
+    # worker processes (trials, ranks, UDF workers) inherit the experiment
+    os.environ["DDLW_EXPERIMENT_ID"] = _active_experiment_id
     return _active_experiment_id
 
 

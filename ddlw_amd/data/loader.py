@@ -139,18 +139,28 @@ class ShardedParquetLoader:
         return [self._rg_index[i] for i in idx]
 
     def _iter_rows(self) -> Iterator[Tuple[bytes, int]]:
-        groups = self._shard_groups()
+        readers = {fi: pq.ParquetFile(f) for fi, f in enumerate(self.files)}
+        if len(self._rg_index) >= self.shard_count:
+            groups = self._shard_groups()
+            row_filter = None
+        else:
+            # fewer row groups than ranks: fall back to row-level round-robin
+            # so every rank still gets a non-empty, disjoint, exhaustive shard
+            groups = list(self._rg_index)
+            row_filter = (self.cur_shard, self.shard_count)
         if not groups:
             return
         epoch = 0
-        readers = {fi: pq.ParquetFile(f) for fi, f in enumerate(self.files)}
         while self.num_epochs is None or epoch < self.num_epochs:
+            row_idx = 0
             for fi, g in groups:
                 tbl = readers[fi].read_row_group(g, columns=[self.content_column, self.label_column])
                 contents = tbl.column(self.content_column).to_pylist()
                 labels = tbl.column(self.label_column).to_pylist()
                 for c, l in zip(contents, labels):
-                    yield c, l
+                    if row_filter is None or row_idx % row_filter[1] == row_filter[0]:
+                        yield c, l
+                    row_idx += 1
             epoch += 1
 
     def _batches_cpu(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:

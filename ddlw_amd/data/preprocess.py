@@ -49,8 +49,8 @@ def preprocess_bytes(content, img_height: int = 224, img_width: int = 224) -> to
     """Main path: decode -> bilinear resize -> [-1,1]. Returns CHW float32
     (torch layout; the HWC->CHW transpose is the only deviation from the
     reference's tensor layout and is internal)."""
-    arr = decode_jpeg(content)
-    t = torch.from_numpy(np.ascontiguousarray(arr)).permute(2, 0, 1).float().unsqueeze(0)
+    arr = np.array(decode_jpeg(content))  # owned, writable copy
+    t = torch.from_numpy(arr).permute(2, 0, 1).float().unsqueeze(0)
     t = torch.nn.functional.interpolate(
         t, size=(img_height, img_width), mode="bilinear", align_corners=False
     )

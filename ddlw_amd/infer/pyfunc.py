@@ -29,11 +29,45 @@ import shutil
 from pathlib import Path
 from typing import Dict, List, Optional, Sequence
 
+import io
+
 import cloudpickle
 import numpy as np
 import yaml
 
 from ..core import tracking
+
+
+def _subimport(name):
+    import importlib
+
+    importlib.import_module(name)
+    import sys as _s
+
+    return _s.modules[name]
+
+
+class _DdlwPickler(cloudpickle.Pickler):
+    """cloudpickle's submodule scan mis-detects ``torch.classes`` as a used
+    submodule whenever a pickled function touches any ``.classes`` attribute
+    (torch registers the lazy torch.classes module in sys.modules); that lazy
+    module is unpicklable. Reduce it — and any torch._classes namespace — to
+    a plain re-import."""
+
+    def reducer_override(self, obj):
+        mod = getattr(type(obj), "__module__", "")
+        if mod == "torch._classes":
+            name = getattr(obj, "__name__", None) or "torch.classes"
+            if not name.startswith("torch"):
+                name = "torch.classes"
+            return (_subimport, (name,))
+        return super().reducer_override(obj)
+
+
+def _dumps(obj) -> bytes:
+    buf = io.BytesIO()
+    _DdlwPickler(buf).dump(obj)
+    return buf.getvalue()
 
 
 class PythonModel:
@@ -75,7 +109,7 @@ def log_model(
             shutil.copy2(src, dst / src.name)
             dst = dst / src.name
         localized[name] = str(dst.relative_to(root))
-    (root / "python_model.pkl").write_bytes(cloudpickle.dumps(python_model))
+    (root / "python_model.pkl").write_bytes(_dumps(python_model))
     (root / "MLmodel").write_text(
         yaml.safe_dump(
             {

@@ -31,6 +31,7 @@ def ddlw_home(tmp_path, monkeypatch):
     monkeypatch.setenv("DDLW_HOME", str(home))
     monkeypatch.delenv("DDLW_TRACKING_URI", raising=False)
     monkeypatch.delenv("DDLW_PARENT_RUN_ID", raising=False)
+    monkeypatch.delenv("DDLW_EXPERIMENT_ID", raising=False)
     import ddlw_amd.core.config as config
     import ddlw_amd.core.tracking as tracking
 
