@@ -1,0 +1,113 @@
+#!/usr/bin/env python3
+"""BASELINE config 5: packaged-model batch inference fan-out — the pyfunc
+predict-UDF fanned to N local GPU workers (the 03_pyfunc path).
+
+Trains nothing: packages a random-init ResNet-50 pyfunc, then measures
+images/sec over a synthetic JPEG batch fanned across workers (decode on CPU
+in each worker, batched bf16 model forward on its GPU).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+from ddlw_amd.core import setup, tracking  # noqa: E402
+from ddlw_amd.core.model_io import log_model  # noqa: E402
+from ddlw_amd.infer import PythonModel, load_model, log_model as log_pyfunc, predict_udf  # noqa: E402
+from ddlw_amd.models import build_resnet50  # noqa: E402
+
+
+class ResNetPyFunc(PythonModel):
+    def load_context(self, context):
+        import json as _json
+
+        from ddlw_amd.core.model_io import load_model as load_torch
+
+        with open(context.artifacts["img_params"]) as f:
+            self.params = _json.load(f)
+        self.model = load_torch(context.artifacts["model"])
+        self.model.eval()
+        if torch.cuda.is_available():
+            self.model = self.model.cuda().to(memory_format=torch.channels_last).bfloat16()
+
+    def predict(self, context, model_input):
+        from ddlw_amd.data.preprocess import preprocess_pil
+
+        h = self.params["img_height"]
+        bs = self.params["batch_size"]
+        outs = []
+        with torch.no_grad():
+            for i in range(0, len(model_input), bs):
+                arrs = np.stack([preprocess_pil(c, h, h) for c in model_input[i : i + bs]])
+                x = torch.from_numpy(arrs).permute(0, 3, 1, 2)
+                if torch.cuda.is_available():
+                    x = x.cuda().to(memory_format=torch.channels_last).bfloat16()
+                outs.append(self.model(x).float().argmax(-1).cpu())
+        return torch.cat(outs).numpy().astype(str)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=None)
+    ap.add_argument("--rows", type=int, default=1000)
+    ap.add_argument("--batch-size", type=int, default=128)
+    args = ap.parse_args()
+    n = args.gpus or max(1, torch.cuda.device_count())
+
+    setup()
+    tracking.set_experiment("bench_infer")
+    with tracking.start_run(run_name="bench_pyfunc") as run:
+        m = build_resnet50(num_classes=1000)
+        model_uri = log_model(m, "model")
+        run.log_dict({"img_height": 224, "batch_size": args.batch_size}, "img_params.json")
+        uri = log_pyfunc(
+            "pyfunc_model",
+            ResNetPyFunc(),
+            artifacts={
+                "img_params": f"runs:/{run.run_id}/img_params.json",
+                "model": model_uri,
+            },
+        )
+
+    from ddlw_amd.data.synthetic import make_synthetic_dataset
+
+    contents, _ = make_synthetic_dataset(args.rows, 224, 224, jpeg=True, seed=1)
+    udf = predict_udf(uri, num_workers=n, gpus=list(range(n)) if torch.cuda.is_available() else [])
+    # warm fan-out (worker model load + MIOpen) then timed
+    udf(contents[: args.batch_size])
+    t0 = time.perf_counter()
+    preds = udf(contents)
+    wall = time.perf_counter() - t0
+    assert len(preds) == args.rows
+    print(
+        json.dumps(
+            {
+                "metric": "images/sec (whole node) ResNet-50 pyfunc batch inference",
+                "value": round(args.rows / wall, 2),
+                "unit": "images/sec",
+                "n_gpus": n if torch.cuda.is_available() else 0,
+                "steps": 1,
+                "warmup": 1,
+                "ms_per_step": round(wall * 1000, 1),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16" if torch.cuda.is_available() else "fp32",
+                "data": "synthetic-jpeg",
+                "config": {"model": "resnet50", "rows": args.rows,
+                           "batch_size": args.batch_size,
+                           "parallelism": f"udf-fanout-{n}"},
+            }
+        )
+    )
+
+
+if __name__ == "__main__":
+    main()

@@ -145,73 +145,122 @@ def load_model(model_uri: str) -> PyFuncModel:
 # --------------------------------------------------------------------------- #
 
 
-def _udf_worker(model_uri: str, rows: List, env: Dict[str, str], q) -> None:
+def _pool_worker(model_uri: str, env: Dict[str, str], task_q, result_q) -> None:
     os.environ.update(env)
     try:
         m = load_model(model_uri)
-        out = m.predict(rows)
-        q.put(("ok", list(np.asarray(out).astype(str))))
+        result_q.put(("ready", os.getpid(), None))
     except Exception:
         import traceback
 
-        q.put(("err", traceback.format_exc()))
+        result_q.put(("err", os.getpid(), traceback.format_exc()))
+        return
+    while True:
+        task = task_q.get()
+        if task is None:
+            return
+        task_id, rows = task
+        try:
+            out = m.predict(rows)
+            result_q.put(("ok", task_id, list(np.asarray(out).astype(str))))
+        except Exception:
+            import traceback
+
+            result_q.put(("err", task_id, traceback.format_exc()))
+
+
+class PredictUDF:
+    """Persistent predict-UDF worker pool (the ``spark_udf`` contract,
+    result_type='string'): each worker process loads the packaged model ONCE,
+    pinned to its own GPU via HIP_VISIBLE_DEVICES, and serves row partitions
+    until closed — like a Spark executor serving Arrow batches."""
+
+    def __init__(self, model_uri: str, num_workers: Optional[int] = None,
+                 gpus: Optional[List[int]] = None):
+        import multiprocessing as mp
+
+        if gpus is None:
+            try:
+                import torch
+
+                gpus = list(range(torch.cuda.device_count())) if torch.cuda.is_available() else []
+            except Exception:
+                gpus = []
+        if num_workers is None:
+            num_workers = max(1, len(gpus)) if gpus else 2
+        self.num_workers = num_workers
+        ctx = mp.get_context("spawn")
+        self._result_q = ctx.Queue()
+        self._task_qs = []
+        self._procs = []
+        base_env = {"DDLW_TRACKING_URI": tracking.get_tracking_uri()}
+        for i in range(num_workers):
+            env = dict(base_env)
+            if gpus:
+                env["HIP_VISIBLE_DEVICES"] = str(gpus[i % len(gpus)])
+            tq = ctx.Queue()
+            p = ctx.Process(
+                target=_pool_worker, args=(model_uri, env, tq, self._result_q),
+                daemon=True,
+            )
+            p.start()
+            self._task_qs.append(tq)
+            self._procs.append(p)
+        ready = 0
+        while ready < num_workers:
+            status, _, payload = self._result_q.get()
+            if status == "err":
+                self.close()
+                raise RuntimeError(f"predict worker failed to load model:\n{payload}")
+            ready += 1
+
+    def __call__(self, rows: Sequence) -> List[str]:
+        rows = list(rows)
+        if not rows:
+            return []
+        n = min(self.num_workers, len(rows))
+        chunk = (len(rows) + n - 1) // n
+        parts = [rows[i * chunk : (i + 1) * chunk] for i in range(n)]
+        for i, part in enumerate(parts):
+            self._task_qs[i].put((i, part))
+        results: List[Optional[List[str]]] = [None] * n
+        errs = []
+        for _ in range(n):
+            status, task_id, payload = self._result_q.get()
+            if status == "ok":
+                results[task_id] = payload
+            else:
+                errs.append(str(payload))
+        if errs:
+            raise RuntimeError("predict_udf worker failure:\n" + "\n".join(errs))
+        out: List[str] = []
+        for part in results:
+            out.extend(part or [])
+        return out
+
+    def close(self) -> None:
+        for tq in self._task_qs:
+            try:
+                tq.put(None)
+            except Exception:
+                pass
+        for p in self._procs:
+            p.join(10)
+            if p.is_alive():
+                p.terminate()
+
+    def __enter__(self) -> "PredictUDF":
+        return self
+
+    def __exit__(self, *exc) -> None:
+        self.close()
 
 
 def predict_udf(
     model_uri: str,
     num_workers: Optional[int] = None,
     gpus: Optional[List[int]] = None,
-):
-    """Return ``udf(rows) -> list[str]`` that fans prediction out over local
-    GPU worker processes (the ``spark_udf`` contract, result_type='string').
-
-    Row partitions are contiguous chunks; each worker loads the packaged
-    model once, pinned to its own GPU via HIP_VISIBLE_DEVICES.
-    """
-    import multiprocessing as mp
-
-    if gpus is None:
-        try:
-            import torch
-
-            gpus = list(range(torch.cuda.device_count())) if torch.cuda.is_available() else []
-        except Exception:
-            gpus = []
-    if num_workers is None:
-        num_workers = max(1, len(gpus)) if gpus else 2
-
-    def udf(rows: Sequence) -> List[str]:
-        rows = list(rows)
-        if not rows:
-            return []
-        n = min(num_workers, len(rows))
-        chunk = (len(rows) + n - 1) // n
-        parts = [rows[i * chunk : (i + 1) * chunk] for i in range(n)]
-        ctx = mp.get_context("spawn")
-        procs = []
-        base_env = {"DDLW_TRACKING_URI": tracking.get_tracking_uri()}
-        for i, part in enumerate(parts):
-            env = dict(base_env)
-            if gpus:
-                env["HIP_VISIBLE_DEVICES"] = str(gpus[i % len(gpus)])
-            q = ctx.SimpleQueue()
-            p = ctx.Process(target=_udf_worker, args=(model_uri, part, env, q), daemon=False)
-            p.start()
-            procs.append((p, q))
-        out: List[str] = []
-        errs = []
-        for p, q in procs:
-            p.join()
-            if not q.empty():
-                status, payload = q.get()
-                if status == "ok":
-                    out.extend(payload)
-                else:
-                    errs.append(payload)
-            else:
-                errs.append(f"worker exited {p.exitcode} with no result")
-        if errs:
-            raise RuntimeError("predict_udf worker failure:\n" + "\n".join(errs))
-        return out
-
-    return udf
+) -> PredictUDF:
+    """Return a callable ``udf(rows) -> list[str]`` backed by a persistent
+    worker pool; call ``.close()`` (or use as a context manager) when done."""
+    return PredictUDF(model_uri, num_workers=num_workers, gpus=gpus)
