@@ -64,6 +64,29 @@ def test_loader_infinite_cycling(ddlw_home):
     conv.delete()
 
 
+@pytest.mark.gpu
+def test_loader_gpu_staging(ddlw_home):
+    """GPU path: pinned-buffer + side-stream H2D staging delivers the same
+    batches as the CPU path."""
+    tbl = _make_table(32, seed=5)
+    conv = make_converter(tbl, row_group_rows=8)
+    dev = torch.device("cuda:0")
+    cpu_batches = []
+    with conv.make_torch_dataset(batch_size=8, num_epochs=1, img_height=16, img_width=16) as loader:
+        cpu_batches = [(i.clone(), l.clone()) for i, l in loader]
+    with conv.make_torch_dataset(
+        batch_size=8, num_epochs=1, img_height=16, img_width=16, device=dev
+    ) as loader:
+        gpu_batches = [(i, l) for i, l in loader]
+        torch.cuda.synchronize()
+    assert len(gpu_batches) == len(cpu_batches)
+    for (ci, cl), (gi, gl) in zip(cpu_batches, gpu_batches):
+        assert gi.is_cuda and gl.is_cuda
+        assert torch.allclose(gi.cpu(), ci)
+        assert torch.equal(gl.cpu(), cl)
+    conv.delete()
+
+
 def test_loader_label_parity(ddlw_home):
     tbl = _make_table(24, seed=3)
     conv = make_converter(tbl, row_group_rows=100)
