@@ -35,7 +35,9 @@ class ResNetPyFunc(PythonModel):
         self.model = load_torch(context.artifacts["model"])
         self.model.eval()
         if torch.cuda.is_available():
-            self.model = self.model.cuda().to(memory_format=torch.channels_last).bfloat16()
+            # fp32 params (BN stats must stay fp32); convs/fc run bf16 under
+            # autocast, activations are bf16 channels_last
+            self.model = self.model.cuda().to(memory_format=torch.channels_last)
 
     def predict(self, context, model_input):
         from ddlw_amd.data.preprocess import preprocess_pil
@@ -49,7 +51,11 @@ class ResNetPyFunc(PythonModel):
                 x = torch.from_numpy(arrs).permute(0, 3, 1, 2)
                 if torch.cuda.is_available():
                     x = x.cuda().to(memory_format=torch.channels_last).bfloat16()
-                outs.append(self.model(x).float().argmax(-1).cpu())
+                    with torch.autocast("cuda", dtype=torch.bfloat16):
+                        logits = self.model(x)
+                else:
+                    logits = self.model(x)
+                outs.append(logits.float().argmax(-1).cpu())
         return torch.cat(outs).numpy().astype(str)
 
 

@@ -90,27 +90,45 @@ __global__ __launch_bounds__(256) void k_bn_stats(
   }
 }
 
-// finalize: reduce partials -> mean/rstd + running-stat update (K4)
+// finalize: reduce partials -> mean/rstd + running-stat update (K4).
+// Block = 32 channels x 8 part-groups (a one-thread-per-channel loop over
+// up to 512 strided partials measured 128 us — pure load latency); each
+// thread sums parts pg::8 (coalesced across the 32 channel lanes), LDS
+// tree over the 8 groups, group 0 finishes the math.
 __global__ __launch_bounds__(256) void k_bn_finalize(
     const float* __restrict__ part_sum, const float* __restrict__ part_sumsq,
     float* __restrict__ mean, float* __restrict__ rstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     long rows, int C, int nparts, float eps, float momentum) {
-  int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
+  const int cl = threadIdx.x & 31;
+  const int pg = threadIdx.x >> 5;
+  const int c = blockIdx.x * 32 + cl;
   float s = 0.f, q = 0.f;
-  for (int p = 0; p < nparts; ++p) {
-    s += part_sum[(long)p * C + c];
-    q += part_sumsq[(long)p * C + c];
+  if (c < C) {
+    for (int p = pg; p < nparts; p += 8) {
+      s += part_sum[(long)p * C + c];
+      q += part_sumsq[(long)p * C + c];
+    }
   }
-  float m = s / (float)rows;
-  float var = fmaxf(q / (float)rows - m * m, 0.f);
-  mean[c] = m;
-  rstd[c] = rsqrtf(var + eps);
-  if (running_mean) {
-    float unbiased = var * (float)rows / (float)(rows > 1 ? rows - 1 : 1);
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
-    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  __shared__ float ls[8][32], lq[8][32];
+  ls[pg][cl] = s;
+  lq[pg][cl] = q;
+  __syncthreads();
+  if (pg == 0 && c < C) {
+    #pragma unroll
+    for (int g = 1; g < 8; ++g) {
+      s += ls[g][cl];
+      q += lq[g][cl];
+    }
+    float m = s / (float)rows;
+    float var = fmaxf(q / (float)rows - m * m, 0.f);
+    mean[c] = m;
+    rstd[c] = rsqrtf(var + eps);
+    if (running_mean) {
+      float unbiased = var * (float)rows / (float)(rows > 1 ? rows - 1 : 1);
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
   }
 }
 
@@ -218,19 +236,33 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
   }
 }
 
-// reduce the bwd partials -> dbeta[C], dgamma[C]
+// reduce the bwd partials -> dbeta[C], dgamma[C] (same geometry as above)
 __global__ __launch_bounds__(256) void k_bn_grad_finalize(
     const float* __restrict__ part_db, const float* __restrict__ part_dg,
     float* __restrict__ dbeta, float* __restrict__ dgamma, int C, int nparts) {
-  int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
+  const int cl = threadIdx.x & 31;
+  const int pg = threadIdx.x >> 5;
+  const int c = blockIdx.x * 32 + cl;
   float b = 0.f, g = 0.f;
-  for (int p = 0; p < nparts; ++p) {
-    b += part_db[(long)p * C + c];
-    g += part_dg[(long)p * C + c];
+  if (c < C) {
+    for (int p = pg; p < nparts; p += 8) {
+      b += part_db[(long)p * C + c];
+      g += part_dg[(long)p * C + c];
+    }
   }
-  dbeta[c] = b;
-  dgamma[c] = g;
+  __shared__ float lb[8][32], lg[8][32];
+  lb[pg][cl] = b;
+  lg[pg][cl] = g;
+  __syncthreads();
+  if (pg == 0 && c < C) {
+    #pragma unroll
+    for (int gg = 1; gg < 8; ++gg) {
+      b += lb[gg][cl];
+      g += lg[gg][cl];
+    }
+    dbeta[c] = b;
+    dgamma[c] = g;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -531,7 +563,7 @@ DDLW_EXPORT int ddlw_bn_finalize(const void* part_sum, const void* part_sumsq,
                                  void* mean, void* rstd, void* rmean, void* rvar,
                                  long rows, int C, float eps, float momentum,
                                  void* stream) {
-  hipLaunchKernelGGL(k_bn_finalize, dim3((C + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(k_bn_finalize, dim3((C + 31) / 32), dim3(256), 0,
                      (hipStream_t)stream, (const float*)part_sum,
                      (const float*)part_sumsq, (float*)mean, (float*)rstd,
                      (float*)rmean, (float*)rvar, rows, C,
@@ -591,7 +623,7 @@ DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* y, const void* x,
 DDLW_EXPORT int ddlw_bn_grad_finalize(const void* part_db, const void* part_dg,
                                       void* dbeta, void* dgamma, long rows,
                                       int C, void* stream) {
-  hipLaunchKernelGGL(k_bn_grad_finalize, dim3((C + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(k_bn_grad_finalize, dim3((C + 31) / 32), dim3(256), 0,
                      (hipStream_t)stream, (const float*)part_db,
                      (const float*)part_dg, (float*)dbeta, (float*)dgamma, C,
                      ddlw_bn_nparts(rows, C));

@@ -108,9 +108,22 @@ class BatchNormAct2d(nn.Module):
         if _use_hip(x, self.num_features):
             if self.training:
                 self.num_batches_tracked += 1
+            # kernels take fp32 parameter/stat pointers; guard against a
+            # model-wide .bfloat16()/.half() cast (silent OOB otherwise)
+            w, b = self.weight, self.bias
+            rm, rv = self.running_mean, self.running_var
+            if w.dtype != torch.float32:
+                w, b = w.float(), b.float()
+            if rm.dtype != torch.float32:
+                if self.training:
+                    raise TypeError(
+                        "BatchNormAct2d training requires fp32 running stats "
+                        f"(got {rm.dtype}); keep BN buffers in fp32"
+                    )
+                rm, rv = rm.float(), rv.float()
             return _BnActFn.apply(
-                x, residual, self.weight, self.bias, self.running_mean,
-                self.running_var, self.training, self.momentum, self.eps, self.relu,
+                x, residual, w, b, rm, rv,
+                self.training, self.momentum, self.eps, self.relu,
             )
         # stock fallback (CPU oracle / non-bf16 path)
         y = F.batch_norm(
