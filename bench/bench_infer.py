@@ -44,10 +44,15 @@ class ResNetPyFunc(PythonModel):
 
         h = self.params["img_height"]
         bs = self.params["batch_size"]
+        from concurrent.futures import ThreadPoolExecutor
+
+        pool = ThreadPoolExecutor(max_workers=8)  # PIL decode releases the GIL
         outs = []
         with torch.no_grad():
             for i in range(0, len(model_input), bs):
-                arrs = np.stack([preprocess_pil(c, h, h) for c in model_input[i : i + bs]])
+                arrs = np.stack(
+                    list(pool.map(lambda c: preprocess_pil(c, h, h), model_input[i : i + bs]))
+                )
                 x = torch.from_numpy(arrs).permute(0, 3, 1, 2)
                 if torch.cuda.is_available():
                     x = x.cuda().to(memory_format=torch.channels_last).bfloat16()

@@ -67,22 +67,30 @@ def bn_stats(x: torch.Tensor, eps: float, momentum: float,
 
 
 def bn_apply(x: torch.Tensor, res: Optional[torch.Tensor], mean, rstd, gamma, beta,
-             relu: bool) -> torch.Tensor:
+             relu: bool):
+    """Returns (y, mask): mask is a uint8 tensor of rows*C/8 relu bits (one
+    byte per 8-channel vector) when relu, else None — backward reads the
+    mask instead of re-reading y."""
     lib = require_lib()
     rows, C = _rows_c(x)
     y = torch.empty_like(x)
+    mask = (
+        torch.empty(rows * (C // 8), dtype=torch.uint8, device=x.device)
+        if relu
+        else None
+    )
     check(
         lib.ddlw_bn_apply(
-            _nhwc_ptr(x), _p(res), _nhwc_ptr(y), _p(mean), _p(rstd), _p(gamma),
-            _p(beta), ctypes.c_long(rows), ctypes.c_int(C),
+            _nhwc_ptr(x), _p(res), _nhwc_ptr(y), _p(mask), _p(mean), _p(rstd),
+            _p(gamma), _p(beta), ctypes.c_long(rows), ctypes.c_int(C),
             ctypes.c_int(1 if relu else 0), ctypes.c_void_p(current_stream_ptr())
         ),
         "bn_apply",
     )
-    return y
+    return y, mask
 
 
-def bn_bwd_reduce(dy, y, x, mean, rstd, relu: bool):
+def bn_bwd_reduce(dy, mask, x, mean, rstd, relu: bool):
     lib = require_lib()
     rows, C = _rows_c(x)
     dev = x.device
@@ -93,7 +101,7 @@ def bn_bwd_reduce(dy, y, x, mean, rstd, relu: bool):
     s = current_stream_ptr()
     check(
         lib.ddlw_bn_bwd_reduce(
-            _nhwc_ptr(dy), _p(y), _nhwc_ptr(x), _p(mean), _p(rstd), _p(part[0]),
+            _nhwc_ptr(dy), _p(mask), _nhwc_ptr(x), _p(mean), _p(rstd), _p(part[0]),
             _p(part[1]), ctypes.c_long(rows), ctypes.c_int(C),
             ctypes.c_int(1 if relu else 0), ctypes.c_void_p(s)
         ),
@@ -109,7 +117,7 @@ def bn_bwd_reduce(dy, y, x, mean, rstd, relu: bool):
     return dbeta, dgamma
 
 
-def bn_bwd_dx(dy, y, x, mean, rstd, gamma, dbeta, dgamma, relu: bool,
+def bn_bwd_dx(dy, mask, x, mean, rstd, gamma, dbeta, dgamma, relu: bool,
               want_dres: bool):
     lib = require_lib()
     rows, C = _rows_c(x)
@@ -117,7 +125,7 @@ def bn_bwd_dx(dy, y, x, mean, rstd, gamma, dbeta, dgamma, relu: bool,
     dres = torch.empty_like(x) if want_dres else None
     check(
         lib.ddlw_bn_bwd_dx(
-            _nhwc_ptr(dy), _p(y), _nhwc_ptr(x), _p(mean), _p(rstd), _p(gamma),
+            _nhwc_ptr(dy), _p(mask), _nhwc_ptr(x), _p(mean), _p(rstd), _p(gamma),
             _p(dbeta), _p(dgamma), _p(dx), _p(dres), ctypes.c_long(rows),
             ctypes.c_int(C), ctypes.c_int(1 if relu else 0),
             ctypes.c_void_p(current_stream_ptr())

@@ -140,7 +140,7 @@ __global__ __launch_bounds__(256) void k_bn_finalize(
 template <int RELU, bool HAS_RES>
 __global__ __launch_bounds__(256) void k_bn_apply(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ res,
-    bf16_t* __restrict__ y,
+    bf16_t* __restrict__ y, unsigned char* __restrict__ mask,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     long rows, int C) {
@@ -152,15 +152,21 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     bf16x8 v, o, rv;
     v.v = *reinterpret_cast<const uint4*>(x + i * 8);
     if (HAS_RES) rv.v = *reinterpret_cast<const uint4*>(res + i * 8);
+    unsigned char mb = 0;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
       const int c = vc * 8 + k;
       float f = (b2f(v.h[k]) - mean[c]) * rstd[c] * gamma[c] + beta[c];
       if (HAS_RES) f += b2f(rv.h[k]);
-      if (RELU) f = fmaxf(f, 0.f);
+      if (RELU) {
+        if (f > 0.f) mb |= (1u << k);
+        f = fmaxf(f, 0.f);
+      }
       o.h[k] = f2b(f);
     }
     *reinterpret_cast<uint4*>(y + i * 8) = o.v;
+    if (RELU) mask[i] = mb;  // 1 byte per 8-channel vector: bwd re-reads
+                             // this instead of the whole y tensor
   }
 }
 
@@ -172,7 +178,7 @@ __global__ __launch_bounds__(256) void k_bn_apply(
 // ---------------------------------------------------------------------------
 template <int RELU>
 __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
-    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
+    const bf16_t* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16_t* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     float* __restrict__ dbeta, float* __restrict__ dgamma,
@@ -194,14 +200,14 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
     const long rstride = (long)gridDim.y * ROWG;
     for (long r = row0; r < rows; r += rstride) {
       const long base = r * C + (long)vec * 8;
-      bf16x8 vdy, vy, vx;
+      bf16x8 vdy, vx;
       vdy.v = *reinterpret_cast<const uint4*>(dy + base);
-      if (RELU) vy.v = *reinterpret_cast<const uint4*>(y + base);
       vx.v = *reinterpret_cast<const uint4*>(x + base);
+      unsigned char mb = RELU ? mask[r * vecC + vec] : 0xff;
       #pragma unroll
       for (int k = 0; k < 8; ++k) {
         float g = b2f(vdy.h[k]);
-        if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
+        if (RELU && !((mb >> k) & 1)) g = 0.f;
         db[k] += g;
         dg[k] += g * (b2f(vx.h[k]) - mn[k]) * rs[k];
       }
@@ -273,7 +279,7 @@ __global__ __launch_bounds__(256) void k_bn_grad_finalize(
 // ---------------------------------------------------------------------------
 template <int RELU, bool HAS_RES>
 __global__ __launch_bounds__(256) void k_bn_bwd_dx(
-    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
+    const bf16_t* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16_t* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma,
@@ -286,15 +292,15 @@ __global__ __launch_bounds__(256) void k_bn_bwd_dx(
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
     const int vc = (int)(i % vecC);
-    bf16x8 vdy, vy, vx, odx, odr;
+    bf16x8 vdy, vx, odx, odr;
     vdy.v = *reinterpret_cast<const uint4*>(dy + i * 8);
-    if (RELU) vy.v = *reinterpret_cast<const uint4*>(y + i * 8);
     vx.v = *reinterpret_cast<const uint4*>(x + i * 8);
+    const unsigned char mb = RELU ? mask[i] : 0xff;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
       const int c = vc * 8 + k;
       float g = b2f(vdy.h[k]);
-      if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
+      if (RELU && !((mb >> k) & 1)) g = 0.f;
       float xhat = (b2f(vx.h[k]) - mean[c]) * rstd[c];
       float d = gamma[c] * rstd[c] * (g - dbeta[c] * invM - xhat * dgamma[c] * invM);
       odx.h[k] = f2b(d);
@@ -572,7 +578,7 @@ DDLW_EXPORT int ddlw_bn_finalize(const void* part_sum, const void* part_sumsq,
 }
 
 DDLW_EXPORT int ddlw_bn_apply(const void* x, const void* res, void* y,
-                              const void* mean, const void* rstd,
+                              void* mask, const void* mean, const void* rstd,
                               const void* gamma, const void* beta, long rows,
                               int C, int relu, void* stream) {
   long nvec = rows * (C >> 3);
@@ -580,27 +586,31 @@ DDLW_EXPORT int ddlw_bn_apply(const void* x, const void* res, void* y,
   if (relu && res)
     hipLaunchKernelGGL((k_bn_apply<1, true>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)x, (const bf16_t*)res, (bf16_t*)y,
+                       (unsigned char*)mask,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)beta, rows, C);
   else if (relu)
     hipLaunchKernelGGL((k_bn_apply<1, false>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)x, nullptr, (bf16_t*)y,
+                       (unsigned char*)mask,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)beta, rows, C);
   else if (res)
     hipLaunchKernelGGL((k_bn_apply<0, true>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)x, (const bf16_t*)res, (bf16_t*)y,
+                       (unsigned char*)mask,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)beta, rows, C);
   else
     hipLaunchKernelGGL((k_bn_apply<0, false>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)x, nullptr, (bf16_t*)y,
+                       (unsigned char*)mask,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)beta, rows, C);
   DDLW_CHECK_LAUNCH();
 }
 
-DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* y, const void* x,
+DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* mask, const void* x,
                                    const void* mean, const void* rstd,
                                    void* dbeta, void* dgamma, long rows, int C,
                                    int relu, void* stream) {
@@ -609,12 +619,12 @@ DDLW_EXPORT int ddlw_bn_bwd_reduce(const void* dy, const void* y, const void* x,
   dim3 grid((vecC + VPB - 1) / VPB, ddlw_bn_nparts(rows, C));
   if (relu)
     hipLaunchKernelGGL((k_bn_bwd_reduce<1>), grid, dim3(256), 0, (hipStream_t)stream,
-                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (float*)dbeta,
                        (float*)dgamma, rows, C);
   else
     hipLaunchKernelGGL((k_bn_bwd_reduce<0>), grid, dim3(256), 0, (hipStream_t)stream,
-                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (float*)dbeta,
                        (float*)dgamma, rows, C);
   DDLW_CHECK_LAUNCH();
@@ -630,7 +640,7 @@ DDLW_EXPORT int ddlw_bn_grad_finalize(const void* part_db, const void* part_dg,
   DDLW_CHECK_LAUNCH();
 }
 
-DDLW_EXPORT int ddlw_bn_bwd_dx(const void* dy, const void* y, const void* x,
+DDLW_EXPORT int ddlw_bn_bwd_dx(const void* dy, const void* mask, const void* x,
                                const void* mean, const void* rstd,
                                const void* gamma, const void* dbeta,
                                const void* dgamma, void* dx, void* dres,
@@ -639,25 +649,25 @@ DDLW_EXPORT int ddlw_bn_bwd_dx(const void* dy, const void* y, const void* x,
   dim3 grid(grid_1d(nvec));
   if (relu && dres)
     hipLaunchKernelGGL((k_bn_bwd_dx<1, true>), grid, dim3(256), 0, (hipStream_t)stream,
-                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
                        (bf16_t*)dres, rows, C);
   else if (relu)
     hipLaunchKernelGGL((k_bn_bwd_dx<1, false>), grid, dim3(256), 0, (hipStream_t)stream,
-                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
                        nullptr, rows, C);
   else if (dres)
     hipLaunchKernelGGL((k_bn_bwd_dx<0, true>), grid, dim3(256), 0, (hipStream_t)stream,
-                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
                        (bf16_t*)dres, rows, C);
   else
     hipLaunchKernelGGL((k_bn_bwd_dx<0, false>), grid, dim3(256), 0, (hipStream_t)stream,
-                       (const bf16_t*)dy, (const bf16_t*)y, (const bf16_t*)x,
+                       (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
                        (const float*)mean, (const float*)rstd, (const float*)gamma,
                        (const float*)dbeta, (const float*)dgamma, (bf16_t*)dx,
                        nullptr, rows, C);

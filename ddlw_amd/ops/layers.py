@@ -57,8 +57,10 @@ class _BnActFn(torch.autograd.Function):
         else:
             mean = running_mean.to(torch.float32)
             rstd = (running_var.to(torch.float32) + eps).rsqrt()
-        y = binding.bn_apply(x, res, mean, rstd, weight, bias, relu)
-        ctx.save_for_backward(x, y, mean, rstd, weight)
+        y, mask = binding.bn_apply(x, res, mean, rstd, weight, bias, relu)
+        if mask is None:
+            mask = x.new_empty(0, dtype=torch.uint8)
+        ctx.save_for_backward(x, mask, mean, rstd, weight)
         ctx.relu = relu
         ctx.has_res = res is not None
         ctx.bn_training = training
@@ -66,18 +68,18 @@ class _BnActFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, mean, rstd, weight = ctx.saved_tensors
+        x, mask, mean, rstd, weight = ctx.saved_tensors
         dy = _cl(dy)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
-        dbeta, dgamma = binding.bn_bwd_reduce(dy, y, x, mean, rstd, ctx.relu)
+        dbeta, dgamma = binding.bn_bwd_reduce(dy, mask, x, mean, rstd, ctx.relu)
         if ctx.bn_training:
             cb, cg = dbeta, dgamma
         else:
             # eval-mode BN: mean/var are constants -> no batch-stat terms
             cb = torch.zeros_like(dbeta)
             cg = torch.zeros_like(dgamma)
-        dx, dres = binding.bn_bwd_dx(dy, y, x, mean, rstd, weight, cb, cg,
+        dx, dres = binding.bn_bwd_dx(dy, mask, x, mean, rstd, weight, cb, cg,
                                      ctx.relu, ctx.has_res)
         return (dx, dres if ctx.has_res else None, dgamma, dbeta,
                 None, None, None, None, None, None)
