@@ -77,6 +77,12 @@ def main() -> None:
     model = build_resnet50(num_classes=num_classes).to(device)
     if use_cuda:
         model = model.to(memory_format=torch.channels_last)
+    if hip_ops:
+        # bf16 weights for convs/fc (no per-step autocast casts); BN params
+        # and stats stay fp32; FusedSGD keeps fp32 masters
+        for m in model.modules():
+            if isinstance(m, (torch.nn.Conv2d, torch.nn.Linear)):
+                m.to(torch.bfloat16)
 
     if hip_ops:
         base_opt = FusedSGD(model.parameters(), lr=0.1 * world, momentum=0.9, weight_decay=1e-4)
@@ -122,8 +128,7 @@ def main() -> None:
             opt.zero_grad(set_to_none=True)
             if hip_ops:
                 x = normalize_u8_bf16(d_img)
-                with torch.autocast("cuda", dtype=amp_dtype):
-                    logits = model(x)
+                logits = model(x)
                 loss = softmax_cross_entropy(logits, d_lab)
             else:
                 x = d_img.to(amp_dtype).mul_(1.0 / 127.5).sub_(1.0)
@@ -161,8 +166,7 @@ def main() -> None:
             opt.zero_grad(set_to_none=False)
             if hip_ops:
                 x = normalize_u8_bf16(d_img)
-                with torch.autocast("cuda", dtype=amp_dtype):
-                    logits = model(x)
+                logits = model(x)
                 loss = softmax_cross_entropy(logits, d_lab)
             else:
                 x = d_img.to(amp_dtype).mul(1.0 / 127.5).sub(1.0)

@@ -28,6 +28,9 @@ def trial(params):
     require_lib()
     torch.manual_seed(0)
     model = build_resnet50(num_classes=1000).to(device).to(memory_format=torch.channels_last)
+    for m in model.modules():
+        if isinstance(m, (torch.nn.Conv2d, torch.nn.Linear)):
+            m.to(torch.bfloat16)
     opt = FusedSGD(model.parameters(), lr=params["lr"], momentum=0.9)
     x = torch.randn(batch, 3, 224, 224, device=device).to(torch.bfloat16).contiguous(
         memory_format=torch.channels_last
@@ -37,8 +40,7 @@ def trial(params):
 
     def one():
         opt.zero_grad(set_to_none=True)
-        with torch.autocast("cuda", dtype=torch.bfloat16):
-            logits = model(x)
+        logits = model(x)
         loss = softmax_cross_entropy(logits, y)
         loss.backward()
         opt.step()

@@ -490,11 +490,12 @@ __global__ __launch_bounds__(256) void k_softmax_ce(
 // master stays fp32).
 // ---------------------------------------------------------------------------
 struct SgdChunk {
-  float* p;
-  const float* g;
-  float* m;
-  bf16_t* p_bf16;  // nullable
+  float* p;        // fp32 master
+  const void* g;   // grad: fp32, or bf16 when (flags & 1)
+  float* m;        // momentum fp32
+  bf16_t* p_bf16;  // nullable bf16 shadow of the weights (the model's param)
   long n;
+  long flags;      // bit0: grad is bf16
 };
 
 __global__ __launch_bounds__(256) void k_fused_sgd(
@@ -502,9 +503,12 @@ __global__ __launch_bounds__(256) void k_fused_sgd(
     float lr, float momentum, float weight_decay, int first_step) {
   for (int ci = blockIdx.y; ci < nchunks; ci += gridDim.y) {
     SgdChunk ch = chunks[ci];
+    const bool g_bf16 = ch.flags & 1;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < ch.n;
          i += (long)gridDim.x * blockDim.x) {
-      float g = ch.g[i] + weight_decay * ch.p[i];
+      float gr = g_bf16 ? b2f(((const bf16_t*)ch.g)[i])
+                        : ((const float*)ch.g)[i];
+      float g = gr + weight_decay * ch.p[i];
       float v = first_step ? g : momentum * ch.m[i] + g;
       ch.m[i] = v;
       float p = ch.p[i] - lr * v;

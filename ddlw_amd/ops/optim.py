@@ -1,11 +1,18 @@
 """Fused SGD optimizer — one multi-tensor kernel per step (north-star
 "fused SGD step"; replaces torch.optim.SGD's per-tensor foreach ops).
 
-Momentum + weight decay, fp32 params/grads/momentum (momentum buffers start
-at zero, so v = mu*v + g reproduces torch.optim.SGD's buf-initialized-to-grad
-first step). The chunk descriptor table (param/grad/momentum pointers +
-sizes) is built once per group and refreshed only if grad storage moves, so
-the steady-state step is one kernel launch per group.
+Momentum + weight decay. Two parameter classes:
+- fp32 params: updated in place (grad fp32);
+- bf16 params (conv/linear weights kept in bf16 so no per-step autocast
+  casts): the optimizer holds an fp32 MASTER copy + momentum; the kernel
+  reads the bf16 grad, updates the master, and writes the bf16 shadow —
+  all in the same pass (mixed-precision SGD exactly like the reference's
+  LossScale-free bf16 recipe).
+
+Momentum buffers start at zero, so v = mu*v + g reproduces torch.optim.SGD's
+buf-initialized-to-grad first step. The chunk descriptor table is built once
+per group and refreshed only if grad storage moves, so the steady-state step
+is one kernel launch per group.
 """
 from __future__ import annotations
 
@@ -23,6 +30,16 @@ class FusedSGD(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self._desc: Dict[int, Tuple[tuple, torch.Tensor, int]] = {}
 
+    def _state_for(self, p: torch.Tensor):
+        st = self.state[p]
+        if "momentum_buffer" not in st:
+            if p.dtype == torch.bfloat16:
+                st["master"] = p.detach().float()
+                st["momentum_buffer"] = torch.zeros_like(st["master"])
+            else:
+                st["momentum_buffer"] = torch.zeros_like(p)
+        return st
+
     def _group_desc(self, gi: int, params: List[torch.Tensor]):
         key = tuple((p.data_ptr(), p.grad.data_ptr(), p.numel()) for p in params)
         cached = self._desc.get(gi)
@@ -31,13 +48,19 @@ class FusedSGD(torch.optim.Optimizer):
         rows = []
         max_numel = 0
         for p in params:
-            st = self.state[p]
-            if "momentum_buffer" not in st:
-                st["momentum_buffer"] = torch.zeros_like(p)
-            rows.append(
-                (p.data_ptr(), p.grad.data_ptr(), st["momentum_buffer"].data_ptr(),
-                 0, p.numel())
-            )
+            st = self._state_for(p)
+            if p.dtype == torch.bfloat16:
+                rows.append(
+                    (st["master"].data_ptr(), p.grad.data_ptr(),
+                     st["momentum_buffer"].data_ptr(), p.data_ptr(), p.numel(),
+                     1 if p.grad.dtype == torch.bfloat16 else 0)
+                )
+            else:
+                assert p.grad.dtype == torch.float32, p.grad.dtype
+                rows.append(
+                    (p.data_ptr(), p.grad.data_ptr(),
+                     st["momentum_buffer"].data_ptr(), 0, p.numel(), 0)
+                )
             max_numel = max(max_numel, p.numel())
         desc = torch.tensor(rows, dtype=torch.int64).to(params[0].device)
         self._desc[gi] = (key, desc, max_numel)
@@ -53,12 +76,17 @@ class FusedSGD(torch.optim.Optimizer):
             if not params[0].is_cuda:
                 # CPU fallback = plain SGD-with-momentum semantics (oracle)
                 for p in params:
-                    st = self.state[p]
-                    if "momentum_buffer" not in st:
-                        st["momentum_buffer"] = torch.zeros_like(p)
-                    g = p.grad + group["weight_decay"] * p
-                    st["momentum_buffer"].mul_(group["momentum"]).add_(g)
-                    p.add_(st["momentum_buffer"], alpha=-group["lr"])
+                    st = self._state_for(p)
+                    if p.dtype == torch.bfloat16:
+                        master = st["master"]
+                        g = p.grad.float() + group["weight_decay"] * master
+                        st["momentum_buffer"].mul_(group["momentum"]).add_(g)
+                        master.add_(st["momentum_buffer"], alpha=-group["lr"])
+                        p.copy_(master.to(torch.bfloat16))
+                    else:
+                        g = p.grad + group["weight_decay"] * p
+                        st["momentum_buffer"].mul_(group["momentum"]).add_(g)
+                        p.add_(st["momentum_buffer"], alpha=-group["lr"])
                 continue
             desc, max_numel = self._group_desc(gi, params)
             binding.fused_sgd(

@@ -185,17 +185,23 @@ class DistributedOptimizer:
         self._param_bucket: Dict[int, Tuple[_Bucket, int]] = {}
         self._hooks = []
         cap = int(bucket_cap_mb * 1024 * 1024)
-        # reverse order ~ backward completion order (last layers first)
-        bucket = _Bucket()
+        # reverse order ~ backward completion order (last layers first);
+        # buckets are per-dtype (bf16 weights + fp32 BN params coexist and
+        # flat buffers must be homogeneous)
+        open_buckets: Dict[torch.dtype, _Bucket] = {}
         for p in reversed(self._params):
             nbytes = p.numel() * p.element_size()
+            bucket = open_buckets.get(p.dtype)
+            if bucket is None:
+                bucket = open_buckets[p.dtype] = _Bucket()
             if bucket.params and bucket.bytes + nbytes > cap:
                 self._buckets.append(bucket)
-                bucket = _Bucket()
+                bucket = open_buckets[p.dtype] = _Bucket()
             bucket.params.append(p)
             bucket.bytes += nbytes
-        if bucket.params:
-            self._buckets.append(bucket)
+        for bucket in open_buckets.values():
+            if bucket.params:
+                self._buckets.append(bucket)
         for b in self._buckets:
             for p in b.params:
                 self._param_bucket[id(p)] = (b, 0)

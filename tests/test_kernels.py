@@ -178,6 +178,31 @@ def test_fused_sgd_matches_torch_sgd():
         assert torch.allclose(a, b, atol=1e-5, rtol=1e-5)
 
 
+def test_fused_sgd_bf16_master_mode():
+    """bf16 params + bf16 grads: fp32 master semantics must match an fp32
+    torch.optim.SGD reference updated with the same (bf16-rounded) grads."""
+    from ddlw_amd.ops.optim import FusedSGD
+
+    torch.manual_seed(23)
+    base = [torch.randn(512, device=_cuda()) for _ in range(2)]
+    p_bf = [b.to(torch.bfloat16).requires_grad_(True) for b in base]
+    p_ref = [b.to(torch.bfloat16).float().requires_grad_(True) for b in base]
+    o1 = FusedSGD(p_bf, lr=0.05, momentum=0.9, weight_decay=1e-4)
+    o2 = torch.optim.SGD(p_ref, lr=0.05, momentum=0.9, weight_decay=1e-4)
+    for step in range(4):
+        gs = [torch.randn(512, device=_cuda()) for _ in range(2)]
+        for a, b, g in zip(p_bf, p_ref, gs):
+            a.grad = g.to(torch.bfloat16)
+            b.grad = g.to(torch.bfloat16).float()
+        o1.step()
+        o2.step()
+    for a, b, in zip(p_bf, p_ref):
+        # bf16 shadow must equal the fp32 reference rounded to bf16
+        assert torch.allclose(a.float(), b.to(torch.bfloat16).float(), atol=1e-2, rtol=1e-2)
+        master = o1.state[a]["master"]
+        assert torch.allclose(master, b, atol=1e-4, rtol=1e-4)
+
+
 def test_normalize_u8_parity():
     from ddlw_amd.ops.layers import normalize_u8_bf16
 
@@ -200,6 +225,9 @@ def test_resnet50_hip_step_runs_and_matches_loss_scale():
 
     torch.manual_seed(17)
     model = build_resnet50(num_classes=1000).to(_cuda()).to(memory_format=torch.channels_last)
+    for m in model.modules():
+        if isinstance(m, (torch.nn.Conv2d, torch.nn.Linear)):
+            m.to(torch.bfloat16)
     opt = FusedSGD(model.parameters(), lr=0.01, momentum=0.9)
     x = _nhwc_bf16(8, 3, 224, 224, seed=19)
     ylab = torch.randint(0, 1000, (8,), device=_cuda())
@@ -207,8 +235,7 @@ def test_resnet50_hip_step_runs_and_matches_loss_scale():
     losses = []
     for _ in range(3):
         opt.zero_grad(set_to_none=True)
-        with torch.autocast("cuda", dtype=torch.bfloat16):
-            logits = model(x)
+        logits = model(x)
         loss = softmax_cross_entropy(logits, ylab)
         loss.backward()
         opt.step()
