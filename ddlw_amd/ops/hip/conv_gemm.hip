@@ -186,19 +186,43 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     cur ^= 1;
   }
 
-  // ---- epilogue: D lane map (16x16): col = lane&15, row = (lane>>4)*4 + q
+  // ---- epilogue: LDS bounce -> fully line-coalesced dwordx4 stores.
+  // The direct D lane map (col = lane&15, row = (lane>>4)*4 + q) emits 2-B
+  // stores touching 4 scattered cache lines per instruction — PMC measured
+  // 67% SQ_WAIT_INST_ANY (store-issue stall) on 1-k-step shapes. Instead
+  // each wave scatters its accs into a private padded LDS tile, then stores
+  // with lane -> (row = lane/CPL, chunk = lane%CPL): every wave-instruction
+  // covers whole consecutive 128-B rows (8 rows x 128 B at WN=64).
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
+  constexpr int WNP = WN + 8;  // pad off the bank power-of-two
+  bf16_t* lC = smem + wave * (WM * WNP);
   #pragma unroll
-  for (int mi = 0; mi < MF; ++mi) {
+  for (int mi = 0; mi < MF; ++mi)
     #pragma unroll
-    for (int ni = 0; ni < NF; ++ni) {
-      int j = tile_n * BN + wc * WN + ni * 16 + d_col;
-      if (j >= K) continue;
+    for (int ni = 0; ni < NF; ++ni)
       #pragma unroll
-      for (int q = 0; q < 4; ++q) {
-        long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
-        if (m < M) y[m * K + j] = f2b(acc[mi][ni][q]);
+      for (int q = 0; q < 4; ++q)
+        lC[(mi * 16 + d_row0 + q) * WNP + ni * 16 + d_col] =
+            f2b(acc[mi][ni][q]);
+  // wave-private region: lgkmcnt ordering suffices, no barrier needed
+  constexpr int CPL = WN / 8;        // 16-B chunks per output row
+  constexpr int RPI = 64 / CPL;      // rows covered per store instruction
+  const int e_row = lane / CPL;      // row within the iteration group
+  const int e_ch = lane % CPL;       // chunk within the row
+  const long m_base = tile_m * BM + wr * WM;
+  const int j_base = tile_n * BN + wc * WN + e_ch * 8;
+  #pragma unroll
+  for (int it = 0; it < WM / RPI; ++it) {
+    const int row = it * RPI + e_row;
+    const long m = m_base + row;
+    uint4 val = *reinterpret_cast<const uint4*>(lC + row * WNP + e_ch * 8);
+    if (m < M) {
+      if (j_base + 8 <= K) {
+        *reinterpret_cast<uint4*>(y + m * K + j_base) = val;
+      } else {
+        const bf16_t* hv = reinterpret_cast<const bf16_t*>(&val);
+        for (int e = 0; e < 8 && j_base + e < K; ++e) y[m * K + j_base + e] = hv[e];
       }
     }
   }
