@@ -31,12 +31,31 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_v;
 #define GLOBAL_AS __attribute__((address_space(1)))
 #define LDS_AS __attribute__((address_space(3)))
 
-template <int BM, int BN>
+// host-precomputed magic division (Granlund-Montgomery): q = (m*M) >> (64)
+// ... here the (32+s)-shift variant; valid for the m < 2^31 row indices.
+// Avoids the ~100-cycle 64-bit div/mod expansion per thread in the prologue
+// (dominant on 1-k-step shapes).
+__device__ __forceinline__ unsigned mdiv(unsigned m, unsigned long long magic,
+                                         unsigned shift) {
+  return (unsigned)(((unsigned long long)m * magic) >> shift);
+}
+
+// hardware RNE float->bf16 (v_cvt path; the integer-rounding f2b is ~5 VALU)
+__device__ __forceinline__ bf16_t f2b_hw(float f) {
+  __bf16 h = (__bf16)f;
+  union { __bf16 h; bf16_t u; } cvt;
+  cvt.h = h;
+  return cvt.u;
+}
+
+template <int BM, int BN, bool EPI_LDS>
 __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
     bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
     int N, int H, int W_, int C, int K, int Ho, int Wo,
-    int R, int S, int stride, int pad, int nwg_swz) {
+    int R, int S, int stride, int pad, int nwg_swz,
+    unsigned long long magic_wo, unsigned shift_wo,
+    unsigned long long magic_ho, unsigned shift_ho) {
   constexpr int BK = 64;
   constexpr int WM = BM / 2, WN = BN / 2;
   constexpr int MF = WM / 16, NF = WN / 16;
@@ -78,10 +97,12 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     long m = tile_m * BM + row;
     a_m[p] = m;
     if (m < M) {
-      int wo = (int)(m % Wo);
-      long t2 = m / Wo;
-      int ho = (int)(t2 % Ho);
-      int n = (int)(t2 / Ho);
+      unsigned mu = (unsigned)m;
+      unsigned q1 = mdiv(mu, magic_wo, shift_wo);     // m / Wo
+      int wo = (int)(mu - q1 * (unsigned)Wo);
+      unsigned n_u = mdiv(q1, magic_ho, shift_ho);    // (m/Wo) / Ho
+      int ho = (int)(q1 - n_u * (unsigned)Ho);
+      int n = (int)n_u;
       a_hb[p] = ho * stride - pad;
       a_wb[p] = wo * stride - pad;
       // source chunk col8' = dest col8 ^ (row&7) = pcol8 ^ prow (swizzle)
@@ -186,15 +207,31 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     cur ^= 1;
   }
 
-  // ---- epilogue: LDS bounce -> fully line-coalesced dwordx4 stores.
-  // The direct D lane map (col = lane&15, row = (lane>>4)*4 + q) emits 2-B
-  // stores touching 4 scattered cache lines per instruction — PMC measured
-  // 67% SQ_WAIT_INST_ANY (store-issue stall) on 1-k-step shapes. Instead
-  // each wave scatters its accs into a private padded LDS tile, then stores
-  // with lane -> (row = lane/CPL, chunk = lane%CPL): every wave-instruction
-  // covers whole consecutive 128-B rows (8 rows x 128 B at WN=64).
+  // ---- epilogue. Two variants:
+  // EPI_LDS=false: direct scalar stores from the D lane map (col = lane&15,
+  //   row = (lane>>4)*4 + q) — measured fastest when the K-loop is long
+  //   (store tail amortized/hidden across resident blocks);
+  // EPI_LDS=true (short K-loop shapes): LDS bounce -> line-coalesced
+  //   dwordx4 stores with lane -> (row = lane/CPL, chunk = lane%CPL): every
+  //   wave-instruction covers whole consecutive 128-B rows.
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
+  if (!EPI_LDS) {
+    #pragma unroll
+    for (int mi = 0; mi < MF; ++mi) {
+      #pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+        if (j >= K) continue;
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
+          if (m < M) y[m * K + j] = f2b_hw(acc[mi][ni][q]);
+        }
+      }
+    }
+    return;
+  }
   constexpr int WNP = WN + 8;  // pad off the bank power-of-two
   bf16_t* lC = smem + wave * (WM * WNP);
   #pragma unroll
@@ -204,7 +241,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
       #pragma unroll
       for (int q = 0; q < 4; ++q)
         lC[(mi * 16 + d_row0 + q) * WNP + ni * 16 + d_col] =
-            f2b(acc[mi][ni][q]);
+            f2b_hw(acc[mi][ni][q]);
   // wave-private region: lgkmcnt ordering suffices, no barrier needed
   constexpr int CPL = WN / 8;        // 16-B chunks per output row
   constexpr int RPI = 64 / CPL;      // rows covered per store instruction
@@ -233,6 +270,16 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
 // ---------------------------------------------------------------------------
 static inline long cdiv(long a, long b) { return (a + b - 1) / b; }
 
+// ceil-magic for q = m/d exact for all m < 2^31 (Granlund-Montgomery):
+// s = ceil(log2 d); magic = ceil(2^(32+s) / d); q = (m*magic) >> (32+s)
+static inline void make_magic(unsigned d, unsigned long long* magic, unsigned* shift) {
+  if (d == 1) { *magic = 1ull << 32; *shift = 32; return; }
+  unsigned s = 0;
+  while ((1ull << s) < d) ++s;
+  *magic = ((1ull << (32 + s)) + d - 1) / d;
+  *shift = 32 + s;
+}
+
 DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
                                     const void* zpage,
                                     int N, int H, int W_, int C, int K,
@@ -243,18 +290,28 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
     return 2;
   }
   long M = (long)N * Ho * Wo;
+  if (M >= (1ll << 31)) {
+    ddlw_set_error("conv_fwd_igemm: M >= 2^31 unsupported");
+    return 2;
+  }
+  unsigned long long mg_wo, mg_ho;
+  unsigned sh_wo, sh_ho;
+  make_magic((unsigned)Wo, &mg_wo, &sh_wo);
+  make_magic((unsigned)Ho, &mg_ho, &sh_ho);
+  const int T = R * S * (C / 64);
   hipStream_t st = (hipStream_t)stream;
-#define LAUNCH(BM, BN)                                                        \
+#define LAUNCH(BM, BN, EPI)                                                   \
   do {                                                                        \
     long grid = cdiv(M, BM) * cdiv(K, BN);                                    \
-    hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN>), dim3((int)grid), dim3(256),\
-                       0, st, (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y, \
-                       (const bf16_t*)zpage, N, H, W_, C, K, Ho, Wo, R, S,    \
-                       stride, pad, (int)grid);                               \
+    hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI>), dim3((int)grid),      \
+                       dim3(256), 0, st, (const bf16_t*)x, (const bf16_t*)w,  \
+                       (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,  \
+                       Wo, R, S, stride, pad, (int)grid, mg_wo, sh_wo, mg_ho, \
+                       sh_ho);                                                \
   } while (0)
-  if (K >= 128) LAUNCH(128, 128);
-  else if (K >= 64) LAUNCH(128, 64);
-  else LAUNCH(128, 32);
+  if (K >= 128) { if (T <= 4) LAUNCH(128, 128, true); else LAUNCH(128, 128, false); }
+  else if (K >= 64) { if (T <= 4) LAUNCH(128, 64, true); else LAUNCH(128, 64, false); }
+  else { if (T <= 4) LAUNCH(128, 32, true); else LAUNCH(128, 32, false); }
 #undef LAUNCH
   DDLW_CHECK_LAUNCH();
 }
