@@ -139,9 +139,8 @@ def maxpool3x3s2_fwd(x: torch.Tensor):
     lib = require_lib()
     n, c, h, w = x.shape
     ho, wo = (h + 1) // 2, (w + 1) // 2
-    y = torch.empty((n, c, ho, wo), dtype=x.dtype, device=x.device).to(
-        memory_format=torch.channels_last
-    )
+    y = torch.empty((n, c, ho, wo), dtype=x.dtype, device=x.device,
+                    memory_format=torch.channels_last)
     argmax = torch.empty(n * ho * wo * c, dtype=torch.uint8, device=x.device)
     check(
         lib.ddlw_maxpool3x3s2_fwd(
@@ -157,9 +156,8 @@ def maxpool3x3s2_bwd(dy: torch.Tensor, argmax: torch.Tensor, in_shape):
     lib = require_lib()
     n, c, h, w = in_shape
     ho, wo = dy.shape[2], dy.shape[3]
-    dx = torch.empty((n, c, h, w), dtype=dy.dtype, device=dy.device).to(
-        memory_format=torch.channels_last
-    )
+    dx = torch.empty((n, c, h, w), dtype=dy.dtype, device=dy.device,
+                     memory_format=torch.channels_last)
     check(
         lib.ddlw_maxpool3x3s2_bwd(
             _nhwc_ptr(dy), _p(argmax), _nhwc_ptr(dx), n, h, w, c, ho, wo,
@@ -182,9 +180,8 @@ def gap_fwd(x: torch.Tensor) -> torch.Tensor:
 def gap_bwd(dy: torch.Tensor, in_shape) -> torch.Tensor:
     lib = require_lib()
     n, c, h, w = in_shape
-    dx = torch.empty((n, c, h, w), dtype=dy.dtype, device=dy.device).to(
-        memory_format=torch.channels_last
-    )
+    dx = torch.empty((n, c, h, w), dtype=dy.dtype, device=dy.device,
+                     memory_format=torch.channels_last)
     check(lib.ddlw_gap_bwd(_p(dy.contiguous()), _nhwc_ptr(dx), n, h * w, c,
                            ctypes.c_void_p(current_stream_ptr())), "gap_bwd")
     return dx
