@@ -97,14 +97,21 @@ def main():
         scale = ref.abs().max().item() + 1e-6
         fwd_ok = err / scale < 5e-2
         dgrad_ok = None
-        if st == 1 and K % 32 == 0:
-            dy = _cl(torch.randn(4, K, Ho, Ho, device=dev).to(torch.bfloat16))
+        dy4 = _cl(torch.randn(4, K, Ho, Ho, device=dev).to(torch.bfloat16))
+        if st == 1 and K % 64 == 0:
             ref_dx = torch.nn.grad.conv2d_input(
-                xs.shape, ws.float(), dy.float(), stride=st, padding=pad
+                xs.shape, ws.float(), dy4.float(), stride=st, padding=pad
             )
-            dx = conv_gemm.conv_dgrad_kernel(dy, ws, xs.shape, pad).float()
+            dx = conv_gemm.conv_dgrad_kernel(dy4, ws, xs.shape, pad).float()
             derr = (dx - ref_dx).abs().max().item() / (ref_dx.abs().max().item() + 1e-6)
             dgrad_ok = derr < 5e-2
+        # wgrad parity (split-K fp32 slabs -> bf16 dW)
+        ref_dw = torch.nn.grad.conv2d_weight(
+            xs.float(), ws.shape, dy4.float(), stride=st, padding=pad
+        )
+        dwh = conv_gemm.conv_wgrad_kernel(dy4, xs, ws.shape, st, pad).float()
+        werr = (dwh - ref_dw).abs().max().item() / (ref_dw.abs().max().item() + 1e-6)
+        wgrad_ok = werr < 5e-2
 
         if args.parity_only:
             report.append({"key": key, "fwd_ok": fwd_ok, "fwd_err": err / scale,
@@ -130,19 +137,34 @@ def main():
             if dgrad_ok
             else None
         )
+        t_stock_w = time_fn(
+            lambda: torch.ops.aten.convolution_backward(
+                dy, x, w, None, [st, st], [pad, pad], [1, 1], False, [0, 0], 1,
+                [False, True, False])[1],
+            args.iters,
+        )
+        t_hip_w = (
+            time_fn(lambda: conv_gemm.conv_wgrad_kernel(dy, x, w.shape, st, pad), args.iters)
+            if wgrad_ok
+            else None
+        )
         ent = {
             "fwd": "hip" if (t_hip_f is not None and t_hip_f < t_stock_f) else "stock",
             "dgrad": "hip" if (t_hip_d is not None and t_hip_d < t_stock_d) else "stock",
+            "wgrad": "hip" if (t_hip_w is not None and t_hip_w < t_stock_w) else "stock",
         }
         table[key] = ent
         rec = {
             "key": key,
             "fwd_ok": fwd_ok,
             "dgrad_ok": dgrad_ok,
+            "wgrad_ok": wgrad_ok,
             "tflops_stock_fwd": round(flops / t_stock_f / 1e9, 1),
             "tflops_hip_fwd": round(flops / t_hip_f / 1e9, 1) if t_hip_f else None,
             "tflops_stock_dgrad": round(flops / t_stock_d / 1e9, 1),
             "tflops_hip_dgrad": round(flops / t_hip_d / 1e9, 1) if t_hip_d else None,
+            "tflops_stock_wgrad": round(flops / t_stock_w / 1e9, 1),
+            "tflops_hip_wgrad": round(flops / t_hip_w / 1e9, 1) if t_hip_w else None,
             "route": ent,
         }
         report.append(rec)

@@ -37,10 +37,10 @@ class Conv2d(nn.Conv2d):
             from . import conv_gemm
 
             if conv_gemm.available(self, x, mode):
-                hip_fwd, hip_dgrad = self._ddlw_route
+                hip_fwd, hip_dgrad, hip_wgrad = self._ddlw_route
                 return conv_gemm.conv2d(
                     x, self.weight, self.bias, self.stride, self.padding,
-                    hip_fwd, hip_dgrad,
+                    hip_fwd, hip_dgrad, hip_wgrad,
                 )
         return F.conv2d(
             x, self.weight.to(x.dtype),
