@@ -40,6 +40,10 @@ def supported_channels(c: int) -> bool:
     return c % 8 == 0
 
 
+import functools
+
+
+@functools.lru_cache(maxsize=512)
 def _nparts(rows: int, C: int) -> int:
     lib = require_lib()
     return int(lib.ddlw_bn_nparts(ctypes.c_long(rows), ctypes.c_int(C)))
