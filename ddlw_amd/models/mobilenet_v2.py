@@ -20,6 +20,8 @@ import torch
 import torch.nn as nn
 
 from ..core.model_io import tag_model
+from ..ops.conv import Conv2d
+from ..ops.layers import BatchNormAct2d, GlobalAvgPool2d
 
 
 class InvertedResidual(nn.Module):
@@ -30,16 +32,16 @@ class InvertedResidual(nn.Module):
         layers = []
         if expand != 1:
             layers += [
-                nn.Conv2d(in_ch, hidden, 1, bias=False),
-                nn.BatchNorm2d(hidden),
+                Conv2d(in_ch, hidden, 1, bias=False),
+                BatchNormAct2d(hidden),
                 nn.ReLU6(inplace=True),
             ]
         layers += [
-            nn.Conv2d(hidden, hidden, 3, stride=stride, padding=1, groups=hidden, bias=False),
-            nn.BatchNorm2d(hidden),
+            Conv2d(hidden, hidden, 3, stride=stride, padding=1, groups=hidden, bias=False),
+            BatchNormAct2d(hidden),
             nn.ReLU6(inplace=True),
-            nn.Conv2d(hidden, out_ch, 1, bias=False),
-            nn.BatchNorm2d(out_ch),
+            Conv2d(hidden, out_ch, 1, bias=False),
+            BatchNormAct2d(out_ch),
         ]
         self.conv = nn.Sequential(*layers)
 
@@ -65,8 +67,8 @@ class MobileNetV2(nn.Module):
     def __init__(self, channels: int = 3):
         super().__init__()
         blocks = [
-            nn.Conv2d(channels, 32, 3, stride=2, padding=1, bias=False),
-            nn.BatchNorm2d(32),
+            Conv2d(channels, 32, 3, stride=2, padding=1, bias=False),
+            BatchNormAct2d(32),
             nn.ReLU6(inplace=True),
         ]
         in_ch = 32
@@ -75,15 +77,15 @@ class MobileNetV2(nn.Module):
                 blocks.append(InvertedResidual(in_ch, c, s if i == 0 else 1, t))
                 in_ch = c
         blocks += [
-            nn.Conv2d(in_ch, 1280, 1, bias=False),
-            nn.BatchNorm2d(1280),
+            Conv2d(in_ch, 1280, 1, bias=False),
+            BatchNormAct2d(1280),
             nn.ReLU6(inplace=True),
         ]
         self.features = nn.Sequential(*blocks)
         for m in self.modules():
-            if isinstance(m, nn.Conv2d):
+            if isinstance(m, (nn.Conv2d, Conv2d)):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out")
-            elif isinstance(m, nn.BatchNorm2d):
+            elif isinstance(m, (nn.BatchNorm2d, BatchNormAct2d)):
                 nn.init.ones_(m.weight)
                 nn.init.zeros_(m.bias)
 
@@ -120,15 +122,15 @@ class TransferModel(nn.Module):
     def __init__(self, channels: int = 3, num_classes: int = 5, dropout: float = 0.5):
         super().__init__()
         self.base = FrozenBase(MobileNetV2(channels))
-        self.global_average_pooling = nn.AdaptiveAvgPool2d(1)
+        self.global_average_pooling = GlobalAvgPool2d()
         self.dropout = nn.Dropout(dropout)
         self.classifier = nn.Linear(1280, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = self.base(x)
-        x = self.global_average_pooling(x).flatten(1)
+        x = self.global_average_pooling(x)  # fused GAP -> (N, C)
         x = self.dropout(x)
-        return self.classifier(x)
+        return self.classifier(x.to(self.classifier.weight.dtype))
 
 
 def build_model(
