@@ -176,8 +176,11 @@ class DistributedOptimizer:
         bucket_cap_mb: float = 32.0,
         average: bool = True,
     ):
+        from ..utils.trace import ChromeTracer
+
         self.optimizer = optimizer
         self.average = average
+        self._tracer = ChromeTracer()  # DDLW_TIMELINE collective events
         self._params: List[torch.nn.Parameter] = [
             p for g in optimizer.param_groups for p in g["params"] if p.requires_grad
         ]
@@ -220,6 +223,13 @@ class DistributedOptimizer:
     def _launch(self, bucket: _Bucket) -> None:
         grads = [p.grad for p in bucket.params]
         flat = torch._utils._flatten_dense_tensors(grads)
+        if self._tracer.enabled:
+            import time as _t
+
+            self._tracer.event(
+                f"allreduce[{flat.numel() * flat.element_size() >> 20}MiB]",
+                "collective", _t.time() * 1e6, 1.0, tid=_rank,
+            )
         # async: NCCL enqueues on the comm stream (overlaps backward); keep
         # the work handle — wait() orders the current stream behind it
         bucket.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)

@@ -23,6 +23,7 @@ import torch.nn.functional as F
 
 from ..core import tracking
 from ..core.model_io import log_model
+from ..utils.trace import ChromeTracer
 from .callbacks import Callback, MetricAverageCallback
 
 _autolog_enabled = False
@@ -171,6 +172,10 @@ class Model:
                 }
             )
 
+        # Horovod-Timeline equivalent: DDLW_TIMELINE=<path> -> chrome trace
+        from ..parallel import api as _api
+
+        tracer = ChromeTracer()
         data_iter = iter(data)
         for epoch in range(epochs):
             for cb in callbacks:
@@ -180,20 +185,22 @@ class Model:
             n = 0
             step = 0
             while steps_per_epoch is None or step < steps_per_epoch:
-                try:
-                    images, labels = next(data_iter)
-                except StopIteration:
-                    if steps_per_epoch is None:
-                        data_iter = iter(data)  # next epoch restarts iterator
-                        break
-                    data_iter = iter(data)
+                with tracer.span(f"data[e{epoch}s{step}]", "data", tid=_api.rank()):
                     try:
                         images, labels = next(data_iter)
                     except StopIteration:
-                        break
+                        if steps_per_epoch is None:
+                            data_iter = iter(data)  # next epoch restarts iterator
+                            break
+                        data_iter = iter(data)
+                        try:
+                            images, labels = next(data_iter)
+                        except StopIteration:
+                            break
                 for cb in callbacks:
                     cb.on_batch_begin(step)
-                logs = self.train_step(images, labels)
+                with tracer.span(f"train_step[e{epoch}s{step}]", "step", tid=_api.rank()):
+                    logs = self.train_step(images, labels)
                 for cb in callbacks:
                     cb.on_batch_end(step, logs)
                 for k, v in logs.items():
@@ -221,6 +228,7 @@ class Model:
 
         for cb in callbacks:
             cb.on_train_end()
+        tracer.save()
         if run is not None and api.rank() == 0:
             log_model(self.module, "model")
         return history

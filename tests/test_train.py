@@ -86,6 +86,21 @@ def test_checkpoint_naming(ddlw_home, tmp_path):
     assert "classifier.weight" in sd
 
 
+def test_timeline_trace(ddlw_home, tmp_path, monkeypatch):
+    """DDLW_TIMELINE -> chrome-trace JSON with step/data spans (the Horovod
+    Timeline equivalent)."""
+    import json
+
+    path = tmp_path / "timeline.json"
+    monkeypatch.setenv("DDLW_TIMELINE", str(path))
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.1)
+    m.fit(_toy_data(), epochs=1, verbose=0)
+    trace = json.loads(path.read_text())
+    names = [e["name"] for e in trace["traceEvents"]]
+    assert any(n.startswith("train_step") for n in names)
+    assert any(n.startswith("data") for n in names)
+
+
 def test_autolog(ddlw_home):
     tracking.set_experiment("autolog")
     autolog(True)
