@@ -7,29 +7,43 @@
 // (blockIdx.z) and write fp32 partial slabs reduced by a second kernel —
 // deterministic, no atomics (same philosophy as the BN reductions).
 //
-// MFMA v_mfma_f32_16x16x32_bf16 with i=k, j=c, kk=m: BOTH fragments are
-// m-major per lane, so both tiles are staged TRANSPOSED through LDS:
-// global reads are natural/coalesced (dy[m][k..k+7], x[m][c..c+7]) and the
-// transpose happens in the ds_write scatter (8 x u16 per 16-B load; each
-// wave-instruction writes 64 consecutive m of one k/c row: conflict-free).
-// LDS rows padded to BMP=72 so the b128 fragment reads hit 16 distinct
-// banks (same 36-dword-stride argument as the fwd kernel).
+// Both MFMA fragments are m-major per lane (kk = m), i.e. TRANSPOSED with
+// respect to the natural [m][k] / [m][c] global layout. v2 avoids the
+// 64-scalar-ds_write transpose of v1 entirely:
+//   - tiles are staged with global_load_lds (async, 1 KiB per
+//     wave-instruction) into a SUBTILED image: consecutive [4 rows][16 cols]
+//     row-major blocks, even-m blocks before odd-m blocks within each 32-m
+//     group — exactly the layout ds_read_b64_tr_b16 gathers conflict-free
+//     (the attention-V recipe: lane l, elem j <- lds[(l&15) + j*16 +
+//     (l>>4)*64]);
+//   - fragments are read with TWO hardware transpose reads each (m 0-3 from
+//     the even blocks, m 4-7 from the odd blocks), inline asm counted by an
+//     explicit lgkmcnt(0) + sched_barrier(0) before the MFMAs
+//     (cdna_hip_programming.md §5.4 rule 18, §5.7 form iii).
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4_v;
+typedef __attribute__((ext_vector_type(2))) unsigned tr64_t;
+
+#define GLOBAL_AS __attribute__((address_space(1)))
+#define LDS_AS __attribute__((address_space(3)))
 
 __device__ __forceinline__ unsigned wg_mdiv(unsigned m, unsigned long long magic,
                                             unsigned shift) {
   return (unsigned)(((unsigned long long)m * magic) >> shift);
 }
 
-#define WG_BM 64    // m rows per k-step
-#define WG_BMP 72   // padded LDS row stride (bf16 elements)
+__device__ __forceinline__ unsigned lds_u32(const bf16_t* p) {
+  return (unsigned)(unsigned long long)(const LDS_AS bf16_t*)p;
+}
+
+#define WG_BM 64  // m rows per k-step
 
 template <int BK, int BC>  // output tile: BK x BC (k x c)
 __global__ __launch_bounds__(256) void k_conv_wgrad(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ x,
+    const bf16_t* __restrict__ zpage,
     float* __restrict__ slab,  // [SPLIT][K][RS*C]
     int N, int H, int W_, int C, int K, int Ho, int Wo,
     int R, int S, int stride, int pad,
@@ -38,9 +52,11 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
     unsigned long long magic_ho, unsigned shift_ho) {
   constexpr int WK = BK / 2, WC = BC / 2;   // per-wave tile (2x2 wave grid)
   constexpr int KF = WK / 16, CF = WC / 16; // fragments
-  // LDS: double-buffered transposed tiles [BK][WG_BM] + [BC][WG_BM]
-  __shared__ __attribute__((aligned(16))) bf16_t smem[2 * (BK + BC) * WG_BMP];
-  constexpr int BUF = (BK + BC) * WG_BMP;
+  constexpr int PA = BK / 16 * 2;           // 1-KiB dy pieces (k16 x m32)
+  constexpr int PB = BC / 16 * 2;           // 1-KiB x pieces
+  constexpr int TILE = WG_BM * BK;          // elements per dy tile
+  constexpr int BUF = WG_BM * (BK + BC);
+  __shared__ __attribute__((aligned(16))) bf16_t smem[2 * BUF];
 
   const long M = (long)N * Ho * Wo;
   const int ctiles = (C + BC - 1) / BC;
@@ -51,17 +67,75 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
   const int sp = blockIdx.z;
 
   const long m0 = (long)sp * m_per_split;
-  const long m1 = min(m0 + m_per_split, M);
+  const long m1 = (m0 + m_per_split < M) ? (m0 + m_per_split) : M;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wr = wave >> 1, wc2 = wave & 1;
 
-  // staging assignment: vec v of a tile -> (mrow = v & 63, c8/k8 = v >> 6);
-  // thread t handles v = t + i*256
-  const int s_m = tid & 63;       // this thread's m row within the step
-  const int s_v0 = tid >> 6;      // first k8/c8 index (stride 4)
+  // ---- staging geometry: within a 1-KiB piece (one (k16, m32) segment of 8
+  // [4][16] blocks, even-m blocks first), this lane's chunk is:
+  //   pos = lane>>3, mr = (lane>>1)&3, half = lane&1
+  //   mb  = pos<4 ? 2*pos : 2*(pos-4)+1   (block order inversion)
+  const int s_pos = lane >> 3;
+  const int s_mr = (lane >> 1) & 3;
+  const int s_half = lane & 1;
+  const int s_mb = (s_pos < 4) ? (2 * s_pos) : (2 * (s_pos - 4) + 1);
+  const int s_mlocal = s_mb * 4 + s_mr;        // m within its 32-m group
+  const int s_koff = s_half * 8;               // k offset within the k16
+
+  // per-(m32) step state: global row pointers + validity for this lane
+  const bf16_t* dyrow[2];
+  const bf16_t* xrow[2];
+  bool dv[2], xv[2];
+
+  auto decompose = [&](long mbase) {
+    #pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      long m = mbase + g * 32 + s_mlocal;
+      dv[g] = m < m1;
+      dyrow[g] = dv[g] ? (dy + m * K) : zpage;
+      bool ok = false;
+      const bf16_t* xr = zpage;
+      if (m < m1) {
+        unsigned mu = (unsigned)m;
+        unsigned q1 = wg_mdiv(mu, magic_wo, shift_wo);
+        int wo = (int)(mu - q1 * (unsigned)Wo);
+        unsigned n_u = wg_mdiv(q1, magic_ho, shift_ho);
+        int ho = (int)(q1 - n_u * (unsigned)Ho);
+        int hh = ho * stride - pad + r;
+        int wwv = wo * stride - pad + s;
+        ok = hh >= 0 && hh < H && wwv >= 0 && wwv < W_;
+        if (ok) xr = x + (((long)(int)n_u * H + hh) * W_ + wwv) * C;
+      }
+      xv[g] = ok;
+      xrow[g] = xr;
+    }
+  };
+
+  // stage one WG_BM-row step into buffer `buf` (all glds; pieces round-robin
+  // over waves; each piece is one (k16, m32) segment)
+  auto stage = [&](int buf) {
+    bf16_t* ldy = smem + buf * BUF;    // dy tile image
+    bf16_t* lx = ldy + TILE;           // x tile image
+    #pragma unroll
+    for (int p = wave; p < PA; p += 4) {
+      const int k16 = p >> 1, g = p & 1;
+      int kk = tile_k * BK + k16 * 16 + s_koff;
+      const bf16_t* src = (dv[g] && kk < K) ? (dyrow[g] + kk) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src, (LDS_AS void*)(ldy + p * 512), 16, 0, 0);
+    }
+    #pragma unroll
+    for (int p = wave; p < PB; p += 4) {
+      const int c16 = p >> 1, g = p & 1;
+      int cc = tile_c * BC + c16 * 16 + s_koff;
+      const bf16_t* src = (xv[g] && cc < C) ? (xrow[g] + cc) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src, (LDS_AS void*)(lx + p * 512), 16, 0, 0);
+    }
+  };
 
   f32x4_v acc[KF][CF];
   #pragma unroll
@@ -69,81 +143,54 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
     #pragma unroll
     for (int b = 0; b < CF; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
 
-  const int fr_row = lane & 15;   // k/c row within a 16-fragment
-  const int fr_m8 = (lane >> 4) * 8;
-
   const long nsteps = (m1 - m0 + WG_BM - 1) / WG_BM;
-
-  // stage one WG_BM-row step into buffer `buf`
-  auto stage = [&](int buf, long mbase) {
-    bf16_t* ldy = smem + buf * BUF;           // [BK][WG_BMP]
-    bf16_t* lx = ldy + BK * WG_BMP;           // [BC][WG_BMP]
-    const long m = mbase + s_m;
-    bool mv = m < m1;
-    // decompose m -> (n, ho, wo) once per step (magic division)
-    int hh = -1, wwv = -1;
-    const bf16_t* xrow = nullptr;
-    if (mv) {
-      unsigned mu = (unsigned)m;
-      unsigned q1 = wg_mdiv(mu, magic_wo, shift_wo);
-      int wo = (int)(mu - q1 * (unsigned)Wo);
-      unsigned n_u = wg_mdiv(q1, magic_ho, shift_ho);
-      int ho = (int)(q1 - n_u * (unsigned)Ho);
-      hh = ho * stride - pad + r;
-      wwv = wo * stride - pad + s;
-      xrow = x + (((long)(int)n_u * H + hh) * W_ + wwv) * C;
-    }
-    const bool xv = mv && hh >= 0 && hh < H && wwv >= 0 && wwv < W_;
-    const bf16_t* dyrow = dy + m * K;
-    // dy tile: BK/8 vectors per m row, strided by 4 over this thread
-    #pragma unroll
-    for (int i = 0; i < BK / 32; ++i) {
-      int k8 = s_v0 + i * 4;
-      int kk = tile_k * BK + k8 * 8;
-      bf16x8 vdy;
-      vdy.v = (mv && kk < K) ? *reinterpret_cast<const uint4*>(dyrow + kk)
-                             : uint4{0, 0, 0, 0};
-      #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        ldy[(k8 * 8 + j) * WG_BMP + s_m] = vdy.h[j];
-    }
-    #pragma unroll
-    for (int i = 0; i < BC / 32; ++i) {
-      int c8 = s_v0 + i * 4;
-      int cc = tile_c * BC + c8 * 8;
-      bf16x8 vx;
-      vx.v = (xv && cc < C) ? *reinterpret_cast<const uint4*>(xrow + cc)
-                            : uint4{0, 0, 0, 0};
-      #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        lx[(c8 * 8 + j) * WG_BMP + s_m] = vx.h[j];
-    }
-  };
-
-  stage(0, m0);
+  decompose(m0);
+  stage(0);
+  if (1 < nsteps) decompose(m0 + WG_BM);
   __syncthreads();
+
   int cur = 0;
   for (long t = 0; t < nsteps; ++t) {
-    if (t + 1 < nsteps) stage(cur ^ 1, m0 + (t + 1) * WG_BM);
+    if (t + 1 < nsteps) {
+      stage(cur ^ 1);
+      if (t + 2 < nsteps) decompose(m0 + (t + 2) * WG_BM);
+    }
     bf16_t* ldy = smem + cur * BUF;
-    bf16_t* lx = ldy + BK * WG_BMP;
+    bf16_t* lx = ldy + TILE;
     #pragma unroll
     for (int mh = 0; mh < 2; ++mh) {  // two 32-m halves of the 64-m step
-      bf16x8_v fk[KF], fc[CF];
+      // fragment loads: 2 hardware transpose reads per fragment
+      tr64_t fk0[KF], fk1[KF], fc0[CF], fc1[CF];
+      // per-lane address = base + lane*8 B: each 16-lane subgroup reads one
+      // 128-B [4][16] block and the HW transpose hands lane l column l&15
       #pragma unroll
-      for (int a = 0; a < KF; ++a)
-        fk[a] = *reinterpret_cast<const bf16x8_v*>(
-            ldy + (wr * WK + a * 16 + fr_row) * WG_BMP + mh * 32 + fr_m8);
+      for (int a = 0; a < KF; ++a) {
+        const int k16 = (wr * WK) / 16 + a;
+        unsigned addr = lds_u32(ldy + ((k16 * 2 + mh) * 8) * 64) + lane * 8;
+        asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(fk0[a]) : "v"(addr));
+        asm volatile("ds_read_b64_tr_b16 %0, %1 offset:512"
+                     : "=v"(fk1[a]) : "v"(addr));
+      }
       #pragma unroll
-      for (int b = 0; b < CF; ++b)
-        fc[b] = *reinterpret_cast<const bf16x8_v*>(
-            lx + (wc2 * WC + b * 16 + fr_row) * WG_BMP + mh * 32 + fr_m8);
+      for (int b = 0; b < CF; ++b) {
+        const int c16 = (wc2 * WC) / 16 + b;
+        unsigned addr = lds_u32(lx + ((c16 * 2 + mh) * 8) * 64) + lane * 8;
+        asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(fc0[b]) : "v"(addr));
+        asm volatile("ds_read_b64_tr_b16 %0, %1 offset:512"
+                     : "=v"(fc1[b]) : "v"(addr));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
       #pragma unroll
       for (int a = 0; a < KF; ++a)
         #pragma unroll
-        for (int b = 0; b < CF; ++b)
+        for (int b = 0; b < CF; ++b) {
+          union { struct { tr64_t lo, hi; } p; bf16x8_v v; } fa, fb;
+          fa.p.lo = fk0[a]; fa.p.hi = fk1[a];
+          fb.p.lo = fc0[b]; fb.p.hi = fc1[b];
           acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              fk[a], fc[b], acc[a][b], 0, 0, 0);
+              fa.v, fb.v, acc[a][b], 0, 0, 0);
+        }
     }
     __syncthreads();
     cur ^= 1;
@@ -194,10 +241,10 @@ static inline void wg_magic(unsigned d, unsigned long long* magic, unsigned* shi
   *shift = 32 + s;
 }
 
-DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, void* slab,
-                                void* dw, int N, int H, int W_, int C, int K,
-                                int Ho, int Wo, int R, int S, int stride,
-                                int pad, int split, void* stream) {
+DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, const void* zpage,
+                                void* slab, void* dw, int N, int H, int W_,
+                                int C, int K, int Ho, int Wo, int R, int S,
+                                int stride, int pad, int split, void* stream) {
   if (C % 8 != 0 || K % 8 != 0) {
     ddlw_set_error("conv_wgrad: C and K must be multiples of 8");
     return 2;
@@ -217,9 +264,10 @@ DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, void* slab,
   do {                                                                        \
     dim3 grid((int)(wg_cdiv(K, BK) * wg_cdiv(C, BC)), R * S, split);          \
     hipLaunchKernelGGL((k_conv_wgrad<BK, BC>), grid, dim3(256), 0, st,        \
-                       (const bf16_t*)dy, (const bf16_t*)x, (float*)slab, N,  \
-                       H, W_, C, K, Ho, Wo, R, S, stride, pad, split,         \
-                       m_per_split, mg_wo, sh_wo, mg_ho, sh_ho);              \
+                       (const bf16_t*)dy, (const bf16_t*)x,                   \
+                       (const bf16_t*)zpage, (float*)slab, N, H, W_, C, K,    \
+                       Ho, Wo, R, S, stride, pad, split, m_per_split, mg_wo,  \
+                       sh_wo, mg_ho, sh_ho);                                  \
   } while (0)
   if (K >= 128 && C >= 128) WLAUNCH(128, 128);
   else if (C >= 128) WLAUNCH(64, 128);
