@@ -216,18 +216,32 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
   }
 }
 
-// reduce fp32 slabs -> bf16 dW (flat [K][RS*C] = channels_last weight grad)
+// reduce fp32 slabs -> bf16 dW (flat [K][RS*C] = channels_last weight grad).
+// Block = 32 elements x 8 split-groups + LDS tree (a one-thread-per-element
+// loop over up to 1024 strided partials is pure load latency — same lesson
+// as the BN finalize kernels).
 __global__ __launch_bounds__(256) void k_wgrad_reduce(
     const float* __restrict__ slab, bf16_t* __restrict__ dw,
     long elems, int split) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < elems;
-       i += (long)gridDim.x * blockDim.x) {
+  const int el = threadIdx.x & 31;
+  const int pg = threadIdx.x >> 5;
+  __shared__ float lds[8][32];
+  for (long i0 = (long)blockIdx.x * 32; i0 < elems; i0 += (long)gridDim.x * 32) {
+    const long i = i0 + el;
     float a = 0.f;
-    for (int p = 0; p < split; ++p) a += slab[(long)p * elems + i];
-    union { float f; unsigned u; } cvt;
-    cvt.f = a;
-    unsigned rb = 0x7FFF + ((cvt.u >> 16) & 1);
-    dw[i] = (bf16_t)((cvt.u + rb) >> 16);
+    if (i < elems)
+      for (int p = pg; p < split; p += 8) a += slab[(long)p * elems + i];
+    lds[pg][el] = a;
+    __syncthreads();
+    if (pg == 0 && i < elems) {
+      #pragma unroll
+      for (int g = 1; g < 8; ++g) a += lds[g][el];
+      union { float f; unsigned u; } cvt;
+      cvt.f = a;
+      unsigned rb = 0x7FFF + ((cvt.u >> 16) & 1);
+      dw[i] = (bf16_t)((cvt.u + rb) >> 16);
+    }
+    __syncthreads();
   }
 }
 
@@ -279,8 +293,8 @@ DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, const void* zpage
     if (err_ != hipSuccess) { ddlw_set_error(hipGetErrorString(err_)); return 1; }
   }
   long elems = (long)K * R * S * C;
-  long g = wg_cdiv(elems, 256);
-  if (g > 2048) g = 2048;
+  long g = wg_cdiv(elems, 32);
+  if (g > 4096) g = 4096;
   hipLaunchKernelGGL(k_wgrad_reduce, dim3((int)g), dim3(256), 0, st,
                      (const float*)slab, (bf16_t*)dw, elems, split);
   DDLW_CHECK_LAUNCH();
