@@ -98,11 +98,11 @@ def main():
         fwd_ok = err / scale < 5e-2
         dgrad_ok = None
         dy4 = _cl(torch.randn(4, K, Ho, Ho, device=dev).to(torch.bfloat16))
-        if st == 1 and K % 64 == 0:
+        if conv_gemm.dgrad_supported(C, K, st, R, pad):
             ref_dx = torch.nn.grad.conv2d_input(
                 xs.shape, ws.float(), dy4.float(), stride=st, padding=pad
             )
-            dx = conv_gemm.conv_dgrad_kernel(dy4, ws, xs.shape, pad).float()
+            dx = conv_gemm.conv_dgrad_kernel(dy4, ws, xs.shape, pad, st).float()
             derr = (dx - ref_dx).abs().max().item() / (ref_dx.abs().max().item() + 1e-6)
             dgrad_ok = derr < 5e-2
         # wgrad parity (split-K fp32 slabs -> bf16 dW)
@@ -133,7 +133,7 @@ def main():
             args.iters,
         )
         t_hip_d = (
-            time_fn(lambda: conv_gemm.conv_dgrad_kernel(dy, w, x.shape, pad), args.iters)
+            time_fn(lambda: conv_gemm.conv_dgrad_kernel(dy, w, x.shape, pad, st), args.iters)
             if dgrad_ok
             else None
         )

@@ -54,6 +54,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
     int N, int H, int W_, int C, int K, int Ho, int Wo,
     int R, int S, int stride, int pad, int nwg_swz,
+    int oH, int oW, int oS,  // output scatter: flat out row = (n*oH + ho*oS)*oW + wo*oS
     unsigned long long magic_wo, unsigned shift_wo,
     unsigned long long magic_ho, unsigned shift_ho) {
   constexpr int BK = 64;
@@ -216,6 +217,15 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   //   wave-instruction covers whole consecutive 128-B rows.
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
+  auto out_row = [&](long m) -> long {
+    if (oS == 1) return m;  // identity scatter (normal fwd)
+    unsigned mu = (unsigned)m;
+    unsigned q1 = mdiv(mu, magic_wo, shift_wo);
+    int wo = (int)(mu - q1 * (unsigned)Wo);
+    unsigned n_u = mdiv(q1, magic_ho, shift_ho);
+    int ho = (int)(q1 - n_u * (unsigned)Ho);
+    return ((long)(int)n_u * oH + (long)ho * oS) * oW + (long)wo * oS;
+  };
   if (!EPI_LDS) {
     #pragma unroll
     for (int mi = 0; mi < MF; ++mi) {
@@ -226,7 +236,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
         #pragma unroll
         for (int q = 0; q < 4; ++q) {
           long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
-          if (m < M) y[m * K + j] = f2b_hw(acc[mi][ni][q]);
+          if (m < M) y[out_row(m) * K + j] = f2b_hw(acc[mi][ni][q]);
         }
       }
     }
@@ -255,11 +265,12 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const long m = m_base + row;
     uint4 val = *reinterpret_cast<const uint4*>(lC + row * WNP + e_ch * 8);
     if (m < M) {
+      const long orow = out_row(m);
       if (j_base + 8 <= K) {
-        *reinterpret_cast<uint4*>(y + m * K + j_base) = val;
+        *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
       } else {
         const bf16_t* hv = reinterpret_cast<const bf16_t*>(&val);
-        for (int e = 0; e < 8 && j_base + e < K; ++e) y[m * K + j_base + e] = hv[e];
+        for (int e = 0; e < 8 && j_base + e < K; ++e) y[orow * K + j_base + e] = hv[e];
       }
     }
   }
@@ -284,7 +295,8 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
                                     const void* zpage,
                                     int N, int H, int W_, int C, int K,
                                     int Ho, int Wo, int R, int S, int stride,
-                                    int pad, void* stream) {
+                                    int pad, int oH, int oW, int oS,
+                                    void* stream) {
   if (C % 64 != 0) {
     ddlw_set_error("conv_fwd_igemm: C must be a multiple of 64");
     return 2;
@@ -306,8 +318,8 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
     hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI>), dim3((int)grid),      \
                        dim3(256), 0, st, (const bf16_t*)x, (const bf16_t*)w,  \
                        (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,  \
-                       Wo, R, S, stride, pad, (int)grid, mg_wo, sh_wo, mg_ho, \
-                       sh_ho);                                                \
+                       Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,   \
+                       sh_wo, mg_ho, sh_ho);                                  \
   } while (0)
   if (K >= 128) { if (T <= 4) LAUNCH(128, 128, true); else LAUNCH(128, 128, false); }
   else if (K >= 64) { if (T <= 4) LAUNCH(128, 64, true); else LAUNCH(128, 64, false); }
