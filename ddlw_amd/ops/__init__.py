@@ -17,7 +17,7 @@ from .layers import (
     softmax_cross_entropy,
     normalize_u8_bf16,
 )
-from .optim import FusedSGD
+from .optim import FusedSGD, FusedAdam
 from .conv import Conv2d
 
 __all__ = [
@@ -31,5 +31,6 @@ __all__ = [
     "softmax_cross_entropy",
     "normalize_u8_bf16",
     "FusedSGD",
+    "FusedAdam",
     "Conv2d",
 ]

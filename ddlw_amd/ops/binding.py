@@ -230,3 +230,21 @@ def normalize_u8(x_u8: torch.Tensor, out_bf16: torch.Tensor):
     assert n % 16 == 0
     check(lib.ddlw_normalize_u8(_p(x_u8), _p(out_bf16), ctypes.c_long(n),
                                 ctypes.c_void_p(current_stream_ptr())), "normalize_u8")
+
+
+def fused_adam(chunk_desc: torch.Tensor, nchunks: int, max_numel: int,
+               lr: float, beta1: float, beta2: float, eps: float,
+               weight_decay: float, bc1: float, bc2: float):
+    """chunk_desc: int64 device tensor [nchunks, 7] of
+    (p_ptr, g_ptr, m_ptr, v_ptr, p_bf16_ptr_or_0, numel, flags)."""
+    lib = require_lib()
+    check(
+        lib.ddlw_fused_adam(
+            _p(chunk_desc), nchunks, ctypes.c_long(max_numel),
+            ctypes.c_float(lr), ctypes.c_float(beta1), ctypes.c_float(beta2),
+            ctypes.c_float(eps), ctypes.c_float(weight_decay),
+            ctypes.c_float(bc1), ctypes.c_float(bc2),
+            ctypes.c_void_p(current_stream_ptr())
+        ),
+        "fused_adam",
+    )

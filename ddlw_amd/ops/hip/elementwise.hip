@@ -518,6 +518,63 @@ __global__ __launch_bounds__(256) void k_fused_sgd(
   }
 }
 
+static inline int grid_1d(long work, int block = 256, int cap = 2048) {
+  long g = (work + block - 1) / block;
+  return (int)(g < cap ? (g > 0 ? g : 1) : cap);
+}
+
+// ---------------------------------------------------------------------------
+// Fused Adam (K10 — the reference's optimizer, Adam 1e-3; P1/02:201).
+// fp32 master path like SGD: m/v/master fp32, optional bf16 param shadow,
+// grads fp32 or bf16. Bias correction folded into the step size host-side?
+// No — step count varies per call; computed in-kernel from step_t.
+// ---------------------------------------------------------------------------
+struct AdamChunk {
+  float* p;        // fp32 master
+  const void* g;   // grad (fp32, or bf16 when flags & 1)
+  float* m;        // first moment
+  float* v;        // second moment
+  bf16_t* p_bf16;  // nullable bf16 shadow
+  long n;
+  long flags;
+};
+
+__global__ __launch_bounds__(256) void k_fused_adam(
+    const AdamChunk* __restrict__ chunks, int nchunks, float lr, float beta1,
+    float beta2, float eps, float weight_decay, float bc1, float bc2) {
+  // bc1 = 1/(1-beta1^t), bc2 = 1/(1-beta2^t) precomputed host-side
+  for (int ci = blockIdx.y; ci < nchunks; ci += gridDim.y) {
+    AdamChunk ch = chunks[ci];
+    const bool g_bf16 = ch.flags & 1;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < ch.n;
+         i += (long)gridDim.x * blockDim.x) {
+      float g = g_bf16 ? b2f(((const bf16_t*)ch.g)[i])
+                       : ((const float*)ch.g)[i];
+      g += weight_decay * ch.p[i];
+      float m = beta1 * ch.m[i] + (1.f - beta1) * g;
+      float v = beta2 * ch.v[i] + (1.f - beta2) * g * g;
+      ch.m[i] = m;
+      ch.v[i] = v;
+      float mh = m * bc1;
+      float vh = v * bc2;
+      float p = ch.p[i] - lr * mh / (sqrtf(vh) + eps);
+      ch.p[i] = p;
+      if (ch.p_bf16) ch.p_bf16[i] = f2b(p);
+    }
+  }
+}
+
+DDLW_EXPORT int ddlw_fused_adam(const void* chunks, int nchunks, long max_numel,
+                                float lr, float beta1, float beta2, float eps,
+                                float weight_decay, float bc1, float bc2,
+                                void* stream) {
+  dim3 grid(grid_1d(max_numel, 256, 512), min(nchunks, 64));
+  hipLaunchKernelGGL(k_fused_adam, grid, dim3(256), 0, (hipStream_t)stream,
+                     (const AdamChunk*)chunks, nchunks, lr, beta1, beta2, eps,
+                     weight_decay, bc1, bc2);
+  DDLW_CHECK_LAUNCH();
+}
+
 // ---------------------------------------------------------------------------
 // Input normalize: uint8 NHWC -> bf16 NHWC, x/127.5 - 1 (the MobileNetV2
 // preprocess_input transform, reference P1/02:126), fused with the H2D'd
@@ -542,11 +599,6 @@ __global__ __launch_bounds__(256) void k_normalize_u8(
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
-static inline int grid_1d(long work, int block = 256, int cap = 2048) {
-  long g = (work + block - 1) / block;
-  return (int)(g < cap ? (g > 0 ? g : 1) : cap);
-}
-
 // shared geometry for the BN-style row reductions; ny (= partial count) must
 // match between the reduce launch and its finalize
 DDLW_EXPORT int ddlw_bn_nparts(long rows, int C) {

@@ -94,3 +94,86 @@ class FusedSGD(torch.optim.Optimizer):
                 group["weight_decay"], False,
             )
         return loss
+
+
+class FusedAdam(torch.optim.Optimizer):
+    """Fused Adam (kernel K10 — the reference's optimizer, Adam 1e-3 at
+    P1/02:201): one multi-tensor kernel per step; fp32 m/v (+ fp32 master for
+    bf16 params with the bf16 shadow updated in the same pass)."""
+
+    def __init__(self, params, lr: float = 1e-3, betas=(0.9, 0.999),
+                 eps: float = 1e-8, weight_decay: float = 0.0):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self._desc: Dict[int, Tuple[tuple, torch.Tensor, int]] = {}
+        self._step_t = 0
+
+    def _state_for(self, p: torch.Tensor):
+        st = self.state[p]
+        if "exp_avg" not in st:
+            if p.dtype == torch.bfloat16:
+                st["master"] = p.detach().float()
+                base = st["master"]
+            else:
+                base = p
+            st["exp_avg"] = torch.zeros_like(base)
+            st["exp_avg_sq"] = torch.zeros_like(base)
+        return st
+
+    def _group_desc(self, gi: int, params: List[torch.Tensor]):
+        key = tuple((p.data_ptr(), p.grad.data_ptr(), p.numel()) for p in params)
+        cached = self._desc.get(gi)
+        if cached is not None and cached[0] == key:
+            return cached[1], cached[2]
+        rows = []
+        max_numel = 0
+        for p in params:
+            st = self._state_for(p)
+            if p.dtype == torch.bfloat16:
+                rows.append(
+                    (st["master"].data_ptr(), p.grad.data_ptr(),
+                     st["exp_avg"].data_ptr(), st["exp_avg_sq"].data_ptr(),
+                     p.data_ptr(), p.numel(),
+                     1 if p.grad.dtype == torch.bfloat16 else 0)
+                )
+            else:
+                rows.append(
+                    (p.data_ptr(), p.grad.data_ptr(), st["exp_avg"].data_ptr(),
+                     st["exp_avg_sq"].data_ptr(), 0, p.numel(), 0)
+                )
+            max_numel = max(max_numel, p.numel())
+        desc = torch.tensor(rows, dtype=torch.int64).to(params[0].device)
+        self._desc[gi] = (key, desc, max_numel)
+        return desc, max_numel
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        self._step_t += 1
+        t = self._step_t
+        for gi, group in enumerate(self.param_groups):
+            params = [p for p in group["params"] if p.grad is not None]
+            if not params:
+                continue
+            b1, b2 = group["betas"]
+            if not params[0].is_cuda:
+                for p in params:
+                    st = self._state_for(p)
+                    master = st.get("master")
+                    tgt = master if master is not None else p
+                    g = p.grad.float() + group["weight_decay"] * tgt
+                    st["exp_avg"].mul_(b1).add_(g, alpha=1 - b1)
+                    st["exp_avg_sq"].mul_(b2).addcmul_(g, g, value=1 - b2)
+                    mh = st["exp_avg"] / (1 - b1 ** t)
+                    vh = st["exp_avg_sq"] / (1 - b2 ** t)
+                    tgt.addcdiv_(mh, vh.sqrt().add_(group["eps"]), value=-group["lr"])
+                    if master is not None:
+                        p.copy_(master.to(torch.bfloat16))
+                continue
+            desc, max_numel = self._group_desc(gi, params)
+            binding.fused_adam(
+                desc, len(params), max_numel, group["lr"], b1, b2,
+                group["eps"], group["weight_decay"],
+                1.0 / (1.0 - b1 ** t), 1.0 / (1.0 - b2 ** t),
+            )
+        return loss

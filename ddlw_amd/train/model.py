@@ -43,13 +43,22 @@ def _make_optimizer(spec, params, lr: Optional[float] = None):
         return spec
     name = str(spec)
     lr = 1e-3 if lr is None else lr
+    key = name.lower()
+    params = list(params)
+    # fused ddlw optimizers on GPU (K10); stock torch otherwise
+    on_gpu = bool(params) and params[0].is_cuda
+    if on_gpu and key in ("adam", "sgd"):
+        from ..ops.optim import FusedAdam, FusedSGD
+
+        if key == "adam":
+            return FusedAdam(params, lr=lr)
+        return FusedSGD(params, lr=lr, momentum=0.0)
     table = {
         "adam": torch.optim.Adam,
         "adadelta": torch.optim.Adadelta,
         "sgd": torch.optim.SGD,
         "adamw": torch.optim.AdamW,
     }
-    key = name.lower()
     if key not in table:
         raise ValueError(f"unknown optimizer {spec!r}")
     return table[key](params, lr=lr)

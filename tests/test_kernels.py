@@ -203,6 +203,44 @@ def test_fused_sgd_bf16_master_mode():
         assert torch.allclose(master, b, atol=1e-4, rtol=1e-4)
 
 
+def test_fused_adam_matches_torch_adam():
+    from ddlw_amd.ops.optim import FusedAdam
+
+    torch.manual_seed(29)
+    p1 = [torch.randn(700, device=_cuda(), requires_grad=True) for _ in range(3)]
+    p2 = [p.detach().clone().requires_grad_(True) for p in p1]
+    o1 = FusedAdam(p1, lr=1e-3, weight_decay=1e-4)
+    o2 = torch.optim.Adam(p2, lr=1e-3, weight_decay=1e-4)
+    for step in range(4):
+        g = [torch.randn(700, device=_cuda()) for _ in range(3)]
+        for a, b, gg in zip(p1, p2, g):
+            a.grad = gg.clone()
+            b.grad = gg.clone()
+        o1.step()
+        o2.step()
+    for a, b in zip(p1, p2):
+        assert torch.allclose(a, b, atol=1e-5, rtol=1e-4)
+
+
+def test_fused_adam_bf16_master():
+    from ddlw_amd.ops.optim import FusedAdam
+
+    torch.manual_seed(31)
+    base = torch.randn(512, device=_cuda())
+    p_bf = base.to(torch.bfloat16).requires_grad_(True)
+    p_ref = base.to(torch.bfloat16).float().requires_grad_(True)
+    o1 = FusedAdam([p_bf], lr=1e-2)
+    o2 = torch.optim.Adam([p_ref], lr=1e-2)
+    for step in range(4):
+        g = torch.randn(512, device=_cuda())
+        p_bf.grad = g.to(torch.bfloat16)
+        p_ref.grad = g.to(torch.bfloat16).float()
+        o1.step()
+        o2.step()
+    assert torch.allclose(o1.state[p_bf]["master"], p_ref, atol=1e-5, rtol=1e-4)
+    assert torch.allclose(p_bf.float(), p_ref.to(torch.bfloat16).float(), atol=1e-2, rtol=1e-2)
+
+
 def test_normalize_u8_parity():
     from ddlw_amd.ops.layers import normalize_u8_bf16
 
