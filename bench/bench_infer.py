@@ -50,8 +50,12 @@ class ResNetPyFunc(PythonModel):
         outs = []
         with torch.no_grad():
             for i in range(0, len(model_input), bs):
+                chunk = list(model_input[i : i + bs])
+                n_real = len(chunk)
+                if n_real < bs:  # pad: a new batch shape would trigger a
+                    chunk += [chunk[-1]] * (bs - n_real)  # fresh MIOpen find
                 arrs = np.stack(
-                    list(pool.map(lambda c: preprocess_pil(c, h, h), model_input[i : i + bs]))
+                    list(pool.map(lambda c: preprocess_pil(c, h, h), chunk))
                 )
                 x = torch.from_numpy(arrs).permute(0, 3, 1, 2)
                 if torch.cuda.is_available():
@@ -60,7 +64,7 @@ class ResNetPyFunc(PythonModel):
                         logits = self.model(x)
                 else:
                     logits = self.model(x)
-                outs.append(logits.float().argmax(-1).cpu())
+                outs.append(logits.float().argmax(-1).cpu()[:n_real])
         return torch.cat(outs).numpy().astype(str)
 
 
