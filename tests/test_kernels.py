@@ -292,3 +292,60 @@ def test_native_lib_is_loaded():
     require_lib()
     maps = open("/proc/self/maps").read()
     assert str(LIB_DIR / LIB_NAME) in maps
+
+
+def test_training_equivalence_hip_vs_stock():
+    """8 training steps of ResNet-50 on a fixed batch: the full ddlw path
+    (bf16 weights, every HIP kernel + FusedSGD) must track the stock fp32
+    path's loss trajectory — an end-to-end numerics gate beyond per-kernel
+    parity."""
+    import os
+
+    from ddlw_amd.models import build_resnet50
+    from ddlw_amd.ops import FusedSGD, softmax_cross_entropy
+
+    torch.manual_seed(99)
+    x32 = torch.randn(16, 3, 64, 64, device=_cuda())
+    y = torch.randint(0, 10, (16,), device=_cuda())
+
+    def run(hip: bool):
+        os.environ["DDLW_DISABLE_HIP_OPS"] = "0" if hip else "1"
+        os.environ["DDLW_CONV"] = "hip" if hip else "stock"
+        try:
+            torch.manual_seed(7)
+            m = build_resnet50(num_classes=10).to(_cuda())
+            losses = []
+            if hip:
+                m = m.to(memory_format=torch.channels_last)
+                for mod in m.modules():
+                    if isinstance(mod, (torch.nn.Conv2d, torch.nn.Linear)):
+                        mod.to(torch.bfloat16)
+                opt = FusedSGD(m.parameters(), lr=0.05, momentum=0.9)
+                xb = x32.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+                for _ in range(8):
+                    opt.zero_grad(set_to_none=True)
+                    loss = softmax_cross_entropy(m(xb), y)
+                    loss.backward()
+                    opt.step()
+                    losses.append(float(loss.detach()))
+            else:
+                opt = torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9)
+                for _ in range(8):
+                    opt.zero_grad(set_to_none=True)
+                    loss = torch.nn.functional.cross_entropy(m(x32), y)
+                    loss.backward()
+                    opt.step()
+                    losses.append(float(loss))
+            return losses
+        finally:
+            os.environ.pop("DDLW_DISABLE_HIP_OPS", None)
+            os.environ.pop("DDLW_CONV", None)
+
+    ref = run(False)
+    hip = run(True)
+    # same init/seed/batch: trajectories must track (bf16 vs fp32 tolerance)
+    assert abs(hip[0] - ref[0]) / max(ref[0], 1e-6) < 0.05, (hip[0], ref[0])
+    # loss must be decreasing on the fixed batch in both paths
+    assert hip[-1] < hip[0] and ref[-1] < ref[0], (hip, ref)
+    for a, b in zip(hip, ref):
+        assert abs(a - b) / max(abs(b), 0.3) < 0.35, (hip, ref)
