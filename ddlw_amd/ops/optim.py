@@ -62,7 +62,8 @@ class FusedSGD(torch.optim.Optimizer):
                      st["momentum_buffer"].data_ptr(), 0, p.numel(), 0)
                 )
             max_numel = max(max_numel, p.numel())
-        desc = torch.tensor(rows, dtype=torch.int64).to(params[0].device)
+        cpu = torch.tensor(rows, dtype=torch.int64).pin_memory()
+        desc = cpu.to(params[0].device, non_blocking=True)
         self._desc[gi] = (key, desc, max_numel)
         return desc, max_numel
 
@@ -142,7 +143,8 @@ class FusedAdam(torch.optim.Optimizer):
                      st["exp_avg_sq"].data_ptr(), 0, p.numel(), 0)
                 )
             max_numel = max(max_numel, p.numel())
-        desc = torch.tensor(rows, dtype=torch.int64).to(params[0].device)
+        cpu = torch.tensor(rows, dtype=torch.int64).pin_memory()
+        desc = cpu.to(params[0].device, non_blocking=True)
         self._desc[gi] = (key, desc, max_numel)
         return desc, max_numel
 
