@@ -64,7 +64,9 @@ class FusedSGD(torch.optim.Optimizer):
             max_numel = max(max_numel, p.numel())
         cpu = torch.tensor(rows, dtype=torch.int64).pin_memory()
         desc = cpu.to(params[0].device, non_blocking=True)
-        self._desc[gi] = (key, desc, max_numel)
+        # keep the pinned source alive until at least the next rebuild (the
+        # async H2D may still be reading it when this function returns)
+        self._desc[gi] = (key, desc, max_numel, cpu)
         return desc, max_numel
 
     @torch.no_grad()
@@ -145,7 +147,9 @@ class FusedAdam(torch.optim.Optimizer):
             max_numel = max(max_numel, p.numel())
         cpu = torch.tensor(rows, dtype=torch.int64).pin_memory()
         desc = cpu.to(params[0].device, non_blocking=True)
-        self._desc[gi] = (key, desc, max_numel)
+        # keep the pinned source alive until at least the next rebuild (the
+        # async H2D may still be reading it when this function returns)
+        self._desc[gi] = (key, desc, max_numel, cpu)
         return desc, max_numel
 
     @torch.no_grad()
