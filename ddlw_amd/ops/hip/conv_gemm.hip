@@ -48,7 +48,7 @@ __device__ __forceinline__ bf16_t f2b_hw(float f) {
   return cvt.u;
 }
 
-template <int BM, int BN, bool EPI_LDS>
+template <int BM, int BN, bool EPI_LDS, int BUFS>
 __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
     bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
@@ -63,8 +63,14 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   constexpr int AP = BM / 32;  // 1-KiB A pieces per wave (8 rows each)
   constexpr int BP = BN / 32;  // 1-KiB B pieces per wave
   constexpr int BUF = (BM + BN) * BK;  // bf16 elements per buffer
+  // BUFS=1 for single-K-step shapes (no pipeline to double-buffer; half the
+  // LDS -> twice the resident blocks on these latency-bound 1x1 layers);
+  // the epilogue LDS bounce needs up to 4*WM*(WN+8) elements.
+  constexpr int SMEM = (BUFS * BUF > 4 * (BM / 2) * (BN / 2 + 8))
+                           ? BUFS * BUF
+                           : 4 * (BM / 2) * (BN / 2 + 8);
 
-  __shared__ __attribute__((aligned(16))) bf16_t smem[2 * BUF];
+  __shared__ __attribute__((aligned(16))) bf16_t smem[SMEM];
 
   const long M = (long)N * Ho * Wo;
   const int tiles_n = (K + BN - 1) / BN;
@@ -176,7 +182,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
 
   int cur = 0;
   for (int t = 0; t < T; ++t) {
-    if (t + 1 < T) {
+    if (BUFS > 1 && t + 1 < T) {
       stage(cur ^ 1, r2, s2, ck2);
       advance();
     }
@@ -205,7 +211,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
               fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
     }
     __syncthreads();
-    cur ^= 1;
+    if (BUFS > 1) cur ^= 1;
   }
 
   // ---- epilogue. Two variants:
@@ -312,18 +318,28 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
   make_magic((unsigned)Ho, &mg_ho, &sh_ho);
   const int T = R * S * (C / 64);
   hipStream_t st = (hipStream_t)stream;
-#define LAUNCH(BM, BN, EPI)                                                   \
+#define LAUNCH(BM, BN, EPI, BUFS)                                             \
   do {                                                                        \
     long grid = cdiv(M, BM) * cdiv(K, BN);                                    \
-    hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI>), dim3((int)grid),      \
-                       dim3(256), 0, st, (const bf16_t*)x, (const bf16_t*)w,  \
-                       (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,  \
-                       Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,   \
-                       sh_wo, mg_ho, sh_ho);                                  \
+    hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS>),                 \
+                       dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x,   \
+                       (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage, N, \
+                       H, W_, C, K, Ho, Wo, R, S, stride, pad, (int)grid, oH, \
+                       oW, oS, mg_wo, sh_wo, mg_ho, sh_ho);                   \
   } while (0)
-  if (K >= 128) { if (T <= 4) LAUNCH(128, 128, true); else LAUNCH(128, 128, false); }
-  else if (K >= 64) { if (T <= 4) LAUNCH(128, 64, true); else LAUNCH(128, 64, false); }
-  else { if (T <= 4) LAUNCH(128, 32, true); else LAUNCH(128, 32, false); }
+  if (K >= 128) {
+    if (T == 1) LAUNCH(128, 128, true, 1);
+    else if (T <= 4) LAUNCH(128, 128, true, 2);
+    else LAUNCH(128, 128, false, 2);
+  } else if (K >= 64) {
+    if (T == 1) LAUNCH(128, 64, true, 1);
+    else if (T <= 4) LAUNCH(128, 64, true, 2);
+    else LAUNCH(128, 64, false, 2);
+  } else {
+    if (T == 1) LAUNCH(128, 32, true, 1);
+    else if (T <= 4) LAUNCH(128, 32, true, 2);
+    else LAUNCH(128, 32, false, 2);
+  }
 #undef LAUNCH
   DDLW_CHECK_LAUNCH();
 }
