@@ -24,11 +24,12 @@ __device__ __forceinline__ float b2f(bf16_t h) {
 }
 
 __device__ __forceinline__ bf16_t f2b(float f) {
-  // round-to-nearest-even bf16
-  union { float f; u32 u; } cvt;
-  cvt.f = f;
-  u32 rounding_bias = 0x7FFF + ((cvt.u >> 16) & 1);
-  return (bf16_t)((cvt.u + rounding_bias) >> 16);
+  // hardware RNE convert (v_cvt path; same rounding as the manual
+  // integer-bias form but ~1 VALU instead of ~5)
+  __bf16 h = (__bf16)f;
+  union { __bf16 h; bf16_t u; } cvt;
+  cvt.h = h;
+  return cvt.u;
 }
 
 __device__ __forceinline__ float warp_reduce_sum(float v) {
