@@ -248,3 +248,26 @@ def fused_adam(chunk_desc: torch.Tensor, nchunks: int, max_numel: int,
         ),
         "fused_adam",
     )
+
+
+def depthwise_fwd(x: torch.Tensor, weight: torch.Tensor, stride: int, padding: int) -> torch.Tensor:
+    """Depthwise conv2d forward (groups == C). weight [C,1,R,S] -> transposed
+    to [R*S][C] host-side so taps read contiguous channel vectors."""
+    lib = require_lib()
+    n, c, h, w = x.shape
+    _, _, r, s = weight.shape
+    ho = (h + 2 * padding - r) // stride + 1
+    wo = (w + 2 * padding - s) // stride + 1
+    w_t = weight.reshape(c, r * s).t().contiguous()  # [R*S][C]
+    if w_t.dtype != torch.bfloat16:
+        w_t = w_t.to(torch.bfloat16)
+    y = torch.empty((n, c, ho, wo), dtype=torch.bfloat16, device=x.device,
+                    memory_format=torch.channels_last)
+    check(
+        lib.ddlw_depthwise_fwd(
+            _nhwc_ptr(x), _p(w_t), _nhwc_ptr(y), n, h, w, c, ho, wo, r, s,
+            stride, padding, ctypes.c_void_p(current_stream_ptr())
+        ),
+        "depthwise_fwd",
+    )
+    return y

@@ -26,14 +26,29 @@ class Conv2d(nn.Conv2d):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         mode = conv_mode()
-        if (
+        hip_ok = (
             mode != "stock"
             and x.is_cuda
             and x.dtype == torch.bfloat16
-            and self.groups == 1
             and self.dilation == (1, 1)
             and os.environ.get("DDLW_DISABLE_HIP_OPS", "0") != "1"
+        )
+        if (
+            hip_ok
+            and self.groups == self.in_channels == self.out_channels
+            and self.in_channels % 8 == 0
+            and self.bias is None
+            and not (x.requires_grad or self.weight.requires_grad)
         ):
+            # depthwise (K2 — MobileNetV2 blocks), inference/frozen path
+            from . import binding
+
+            st = self.stride[0] if isinstance(self.stride, tuple) else self.stride
+            pd = self.padding[0] if isinstance(self.padding, tuple) else self.padding
+            return binding.depthwise_fwd(
+                x.contiguous(memory_format=torch.channels_last), self.weight, st, pd
+            )
+        if hip_ok and self.groups == 1:
             from . import conv_gemm
 
             if conv_gemm.available(self, x, mode):

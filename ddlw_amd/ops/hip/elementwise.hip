@@ -793,3 +793,61 @@ DDLW_EXPORT int ddlw_normalize_u8(const void* x, void* y, long n, void* stream) 
                      n / 16);
   DDLW_CHECK_LAUNCH();
 }
+
+// ---------------------------------------------------------------------------
+// Depthwise conv2d forward (K2 — MobileNetV2's 3x3 depthwise blocks), NHWC
+// bf16. Memory-bound: one thread per 8-channel vector of one output pixel,
+// weights pre-transposed host-side to [R*S][C] so the per-tap weight read is
+// the same contiguous 16-B vector shape as the activation read.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_depthwise_fwd(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w_t,
+    bf16_t* __restrict__ y, int N, int H, int W_, int C, int Ho, int Wo,
+    int R, int S, int stride, int pad) {
+  const int vecC = C >> 3;
+  const long total = (long)N * Ho * Wo * vecC;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int vc = (int)(i % vecC);
+    long t = i / vecC;
+    int wo = (int)(t % Wo); t /= Wo;
+    int ho = (int)(t % Ho); t /= Ho;
+    int n = (int)t;
+    const int hb = ho * stride - pad, wb = wo * stride - pad;
+    float acc[8] = {0};
+    for (int r = 0; r < R; ++r) {
+      int h = hb + r;
+      if (h < 0 || h >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        int ww = wb + s;
+        if (ww < 0 || ww >= W_) continue;
+        bf16x8 vx, vw;
+        vx.v = *reinterpret_cast<const uint4*>(
+            x + (((long)n * H + h) * W_ + ww) * C + (long)vc * 8);
+        vw.v = *reinterpret_cast<const uint4*>(
+            w_t + ((long)r * S + s) * C + (long)vc * 8);
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) acc[k] += b2f(vx.h[k]) * b2f(vw.h[k]);
+      }
+    }
+    bf16x8 o;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) o.h[k] = f2b(acc[k]);
+    *reinterpret_cast<uint4*>(y + i * 8) = o.v;
+  }
+}
+
+DDLW_EXPORT int ddlw_depthwise_fwd(const void* x, const void* w_t, void* y,
+                                   int N, int H, int W_, int C, int Ho, int Wo,
+                                   int R, int S, int stride, int pad,
+                                   void* stream) {
+  if (C % 8 != 0) {
+    ddlw_set_error("depthwise_fwd: C must be a multiple of 8");
+    return 2;
+  }
+  long total = (long)N * Ho * Wo * (C >> 3);
+  hipLaunchKernelGGL(k_depthwise_fwd, dim3(grid_1d(total)), dim3(256), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (const bf16_t*)w_t,
+                     (bf16_t*)y, N, H, W_, C, Ho, Wo, R, S, stride, pad);
+  DDLW_CHECK_LAUNCH();
+}

@@ -114,3 +114,20 @@ def test_autolog(ddlw_home):
         assert (r.dir / "artifacts" / "model" / "state_dict.pt").exists()
     finally:
         autolog(False)
+
+
+def test_checkpoint_resume(ddlw_home, tmp_path):
+    """resume from a checkpoint-{epoch}.ckpt file and continue
+    training (the broadcast callback covers rank parity in the DP case)."""
+    m1 = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.05)
+    ckpt = tmp_path / "ck" / "checkpoint-{epoch}.ckpt"
+    m1.fit(_toy_data(), epochs=2, callbacks=[ModelCheckpoint(str(ckpt))], verbose=0)
+
+    # fresh model resumes from epoch-2 weights
+    m2 = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.05)
+    sd = torch.load(tmp_path / "ck" / "checkpoint-2.ckpt", weights_only=True)
+    m2.module.load_state_dict(sd)
+    for (k1, v1), (k2, v2) in zip(m1.module.state_dict().items(), m2.module.state_dict().items()):
+        assert torch.allclose(v1, v2), k1
+    hist = m2.fit(_toy_data(), epochs=1, verbose=0)
+    assert len(hist.history["loss"]) == 1
