@@ -39,7 +39,7 @@ def test_conv_fwd_kernel_parity(shape):
     assert ((y - ref).abs().max() / scale).item() < 5e-2
 
 
-@pytest.mark.parametrize("shape", [s for s in SHAPES if s[6] == 1])
+@pytest.mark.parametrize("shape", [s for s in SHAPES if s[6] == 1] + [(14, 14, 64, 64, 3, 3, 2)])
 def test_conv_dgrad_kernel_parity(shape):
     from ddlw_amd.ops import conv_gemm
 
@@ -49,7 +49,7 @@ def test_conv_dgrad_kernel_parity(shape):
     w = _cl(torch.randn(K, C, R, S, device=_cuda()).to(torch.bfloat16))
     Ho = (H + 2 * pad - R) // st + 1
     dy = _cl(torch.randn(3, K, Ho, Ho, device=_cuda()).to(torch.bfloat16))
-    dx = conv_gemm.conv_dgrad_kernel(dy, w, (3, C, H, W), pad).float()
+    dx = conv_gemm.conv_dgrad_kernel(dy, w, (3, C, H, W), pad, st).float()
     ref = torch.nn.grad.conv2d_input((3, C, H, W), w.float(), dy.float(), stride=st, padding=pad)
     scale = ref.abs().max() + 1e-6
     assert ((dx - ref).abs().max() / scale).item() < 5e-2
