@@ -27,6 +27,7 @@
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4_v;
+typedef __attribute__((ext_vector_type(16))) float f32x16_v;
 
 #define GLOBAL_AS __attribute__((address_space(1)))
 #define LDS_AS __attribute__((address_space(3)))
@@ -156,14 +157,30 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     }
   };
 
-  f32x4_v acc[MF][NF];
-  #pragma unroll
-  for (int mi = 0; mi < MF; ++mi)
+  // MFMA shape: 32x32x16 when the per-wave tile is >=32 in both dims
+  // (2382 vs 2075 TF/s ubench ceiling over 16x16x32); 16x16x32 otherwise.
+  constexpr bool M32 = (WM >= 32) && (WN >= 32);
+  constexpr int MF2 = M32 ? WM / 32 : 1, NF2 = M32 ? WN / 32 : 1;
+  f32x4_v acc[M32 ? 1 : MF][M32 ? 1 : NF];
+  f32x16_v acc2[MF2][NF2];
+  if constexpr (M32) {
     #pragma unroll
-    for (int ni = 0; ni < NF; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+    for (int mi = 0; mi < MF2; ++mi)
+      #pragma unroll
+      for (int ni = 0; ni < NF2; ++ni)
+        #pragma unroll
+        for (int e = 0; e < 16; ++e) acc2[mi][ni][e] = 0.f;
+  } else {
+    #pragma unroll
+    for (int mi = 0; mi < MF; ++mi)
+      #pragma unroll
+      for (int ni = 0; ni < NF; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+  }
 
   const int fr_row = lane & 15;
   const int fr_c8 = lane >> 4;  // col8 base per k-half: kh*4 + (lane>>4)
+  const int fr_row32 = lane & 31;
+  const int fr_q8 = lane >> 5;  // k-eighth within a 16-K quarter
 
   const int csteps = C / BK;
   const int T = R * S * csteps;
@@ -188,27 +205,54 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     }
     bf16_t* lA = smem + cur * BUF;
     bf16_t* lB = lA + BM * BK;
-    #pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
-      bf16x8_v fa[MF], fb[NF];
+    if constexpr (M32) {
+      // four 16-K quarters per BK=64; A/B fragment per lane: 8 bf16 at
+      // [row = lane&31][k = kq*16 + (lane>>5)*8]
       #pragma unroll
-      for (int mi = 0; mi < MF; ++mi) {
-        int row = wr * WM + mi * 16 + fr_row;
-        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
-        fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
-      }
-      #pragma unroll
-      for (int ni = 0; ni < NF; ++ni) {
-        int row = wc * WN + ni * 16 + fr_row;
-        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
-        fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
-      }
-      #pragma unroll
-      for (int mi = 0; mi < MF; ++mi)
+      for (int kq = 0; kq < 4; ++kq) {
+        bf16x8_v fa[MF2], fb[NF2];
         #pragma unroll
-        for (int ni = 0; ni < NF; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
+        for (int mi = 0; mi < MF2; ++mi) {
+          int row = wr * WM + mi * 32 + fr_row32;
+          int c8 = (kq * 2 + fr_q8) ^ (row & 7);
+          fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
+        }
+        #pragma unroll
+        for (int ni = 0; ni < NF2; ++ni) {
+          int row = wc * WN + ni * 32 + fr_row32;
+          int c8 = (kq * 2 + fr_q8) ^ (row & 7);
+          fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
+        }
+        #pragma unroll
+        for (int mi = 0; mi < MF2; ++mi)
+          #pragma unroll
+          for (int ni = 0; ni < NF2; ++ni)
+            acc2[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                fa[mi], fb[ni], acc2[mi][ni], 0, 0, 0);
+      }
+    } else {
+      #pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+        bf16x8_v fa[MF], fb[NF];
+        #pragma unroll
+        for (int mi = 0; mi < MF; ++mi) {
+          int row = wr * WM + mi * 16 + fr_row;
+          int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+          fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
+        }
+        #pragma unroll
+        for (int ni = 0; ni < NF; ++ni) {
+          int row = wc * WN + ni * 16 + fr_row;
+          int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+          fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
+        }
+        #pragma unroll
+        for (int mi = 0; mi < MF; ++mi)
+          #pragma unroll
+          for (int ni = 0; ni < NF; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
+      }
     }
     __syncthreads();
     if (BUFS > 1) cur ^= 1;
