@@ -276,17 +276,37 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     int ho = (int)(q1 - n_u * (unsigned)Ho);
     return ((long)(int)n_u * oH + (long)ho * oS) * oW + (long)wo * oS;
   };
+  // 32x32 D map: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const int d_col32 = lane & 31;
+  const int d_rbase32 = 4 * (lane >> 5);
   if (!EPI_LDS) {
-    #pragma unroll
-    for (int mi = 0; mi < MF; ++mi) {
+    if constexpr (M32) {
       #pragma unroll
-      for (int ni = 0; ni < NF; ++ni) {
-        int j = tile_n * BN + wc * WN + ni * 16 + d_col;
-        if (j >= K) continue;
+      for (int mi = 0; mi < MF2; ++mi) {
         #pragma unroll
-        for (int q = 0; q < 4; ++q) {
-          long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
-          if (m < M) y[out_row(m) * K + j] = f2b_hw(acc[mi][ni][q]);
+        for (int ni = 0; ni < NF2; ++ni) {
+          int j = tile_n * BN + wc * WN + ni * 32 + d_col32;
+          if (j >= K) continue;
+          #pragma unroll
+          for (int reg = 0; reg < 16; ++reg) {
+            int row = (reg & 3) + 8 * (reg >> 2) + d_rbase32;
+            long m = tile_m * BM + wr * WM + mi * 32 + row;
+            if (m < M) y[out_row(m) * K + j] = f2b_hw(acc2[mi][ni][reg]);
+          }
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int mi = 0; mi < MF; ++mi) {
+        #pragma unroll
+        for (int ni = 0; ni < NF; ++ni) {
+          int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+          if (j >= K) continue;
+          #pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
+            if (m < M) y[out_row(m) * K + j] = f2b_hw(acc[mi][ni][q]);
+          }
         }
       }
     }
@@ -294,14 +314,27 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   }
   constexpr int WNP = WN + 8;  // pad off the bank power-of-two
   bf16_t* lC = smem + wave * (WM * WNP);
-  #pragma unroll
-  for (int mi = 0; mi < MF; ++mi)
+  if constexpr (M32) {
     #pragma unroll
-    for (int ni = 0; ni < NF; ++ni)
+    for (int mi = 0; mi < MF2; ++mi)
       #pragma unroll
-      for (int q = 0; q < 4; ++q)
-        lC[(mi * 16 + d_row0 + q) * WNP + ni * 16 + d_col] =
-            f2b_hw(acc[mi][ni][q]);
+      for (int ni = 0; ni < NF2; ++ni)
+        #pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          int row = (reg & 3) + 8 * (reg >> 2) + d_rbase32;
+          lC[(mi * 32 + row) * WNP + ni * 32 + d_col32] =
+              f2b_hw(acc2[mi][ni][reg]);
+        }
+  } else {
+    #pragma unroll
+    for (int mi = 0; mi < MF; ++mi)
+      #pragma unroll
+      for (int ni = 0; ni < NF; ++ni)
+        #pragma unroll
+        for (int q = 0; q < 4; ++q)
+          lC[(mi * 16 + d_row0 + q) * WNP + ni * 16 + d_col] =
+              f2b_hw(acc[mi][ni][q]);
+  }
   // wave-private region: lgkmcnt ordering suffices, no barrier needed
   constexpr int CPL = WN / 8;        // 16-B chunks per output row
   constexpr int RPI = 64 / CPL;      // rows covered per store instruction
