@@ -24,6 +24,7 @@
 //     v_mfma_f32_16x16x32_bf16, two K-halves per BK=64 step;
 //   - XCD-aware bijective workgroup swizzle (L2 tile locality).
 #include "common.h"
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4_v;
@@ -49,7 +50,7 @@ __device__ __forceinline__ bf16_t f2b_hw(float f) {
   return cvt.u;
 }
 
-template <int BM, int BN, bool EPI_LDS, int BUFS>
+template <int BM, int BN, bool EPI_LDS, int BUFS, bool M32EN = true>
 __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
     bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
@@ -159,7 +160,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
 
   // MFMA shape: 32x32x16 when the per-wave tile is >=32 in both dims
   // (2382 vs 2075 TF/s ubench ceiling over 16x16x32); 16x16x32 otherwise.
-  constexpr bool M32 = (WM >= 32) && (WN >= 32);
+  constexpr bool M32 = M32EN && (WM >= 32) && (WN >= 32);
   constexpr int MF2 = M32 ? WM / 32 : 1, NF2 = M32 ? WN / 32 : 1;
   f32x4_v acc[M32 ? 1 : MF][M32 ? 1 : NF];
   f32x16_v acc2[MF2][NF2];
@@ -395,14 +396,26 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
   make_magic((unsigned)Ho, &mg_ho, &sh_ho);
   const int T = R * S * (C / 64);
   hipStream_t st = (hipStream_t)stream;
+  static int mfma_pref = -1;
+  if (mfma_pref < 0) {
+    const char* e = getenv("DDLW_CONV_MFMA");
+    mfma_pref = (e && e[0] == '1' && e[1] == '6') ? 16 : 32;
+  }
 #define LAUNCH(BM, BN, EPI, BUFS)                                             \
   do {                                                                        \
     long grid = cdiv(M, BM) * cdiv(K, BN);                                    \
-    hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS>),                 \
-                       dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x,   \
-                       (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage, N, \
-                       H, W_, C, K, Ho, Wo, R, S, stride, pad, (int)grid, oH, \
-                       oW, oS, mg_wo, sh_wo, mg_ho, sh_ho);                   \
+    if (mfma_pref == 16)                                                      \
+      hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, false>),        \
+                         dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
+                         (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
+                         N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
+                         (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho);  \
+    else                                                                      \
+      hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, true>),         \
+                         dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
+                         (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
+                         N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
+                         (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho);  \
   } while (0)
   if (K >= 128) {
     if (T == 1) LAUNCH(128, 128, true, 1);
