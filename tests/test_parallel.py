@@ -101,3 +101,39 @@ def _failure_worker():
 def test_runner_worker_failure_detected():
     with pytest.raises(RuntimeError):
         Runner(np=2, timeout_s=60).run(_failure_worker)
+
+
+def _distopt_fused_worker(seed):
+    """DistributedOptimizer wrapping FusedSGD (the bench.py N>1 config),
+    CPU-fallback semantics vs a single-process oracle."""
+    from ddlw_amd.ops.optim import FusedSGD
+    from ddlw_amd.parallel import api
+
+    torch.manual_seed(seed)
+    m = torch.nn.Linear(8, 4)
+    api.broadcast_parameters(m, root_rank=0)
+    opt = api.DistributedOptimizer(
+        FusedSGD(m.parameters(), lr=0.1, momentum=0.9), bucket_cap_mb=0.0001
+    )
+    g = torch.Generator().manual_seed(300 + api.rank())
+    x = torch.randn(4, 8, generator=g)
+    for _ in range(2):
+        opt.zero_grad()
+        m(x).sum().backward()
+        opt.step()
+    return {k: v.detach().clone() for k, v in m.state_dict().items()}
+
+
+def test_distributed_fused_sgd_matches_oracle():
+    out = Runner(np=2, timeout_s=120).run(_distopt_fused_worker, seed=11)
+
+    torch.manual_seed(11)
+    m = torch.nn.Linear(8, 4)
+    opt = torch.optim.SGD(m.parameters(), lr=0.1, momentum=0.9)
+    xs = [torch.randn(4, 8, generator=torch.Generator().manual_seed(300 + r)) for r in range(2)]
+    for _ in range(2):
+        opt.zero_grad()
+        (sum(m(x).sum() for x in xs) / 2).backward()
+        opt.step()
+    for k, v in m.state_dict().items():
+        assert torch.allclose(out[k], v, atol=1e-5), k
