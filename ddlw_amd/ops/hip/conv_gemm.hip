@@ -353,8 +353,15 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
       if (j_base + 8 <= K) {
         *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
       } else {
-        const bf16_t* hv = reinterpret_cast<const bf16_t*>(&val);
-        for (int e = 0; e < 8 && j_base + e < K; ++e) y[orow * K + j_base + e] = hv[e];
+        // static component extraction (a reinterpret pointer into `val`
+        // forces the register to scratch — rule 20)
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          if (j_base + e < K) {
+            unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
+            y[orow * K + j_base + e] = (bf16_t)(wd >> ((e & 1) * 16));
+          }
+        }
       }
     }
   }
