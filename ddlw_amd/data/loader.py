@@ -30,7 +30,6 @@ from concurrent.futures import ThreadPoolExecutor
 from pathlib import Path
 from typing import Callable, Iterator, List, Optional, Tuple
 
-import numpy as np
 import pyarrow as pa
 import pyarrow.parquet as pq
 import torch

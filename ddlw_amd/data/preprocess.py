@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import ast
 import io
-from typing import Iterable, Tuple
+from typing import Iterable
 
 import numpy as np
 import torch

@@ -4,7 +4,7 @@ from __future__ import annotations
 
 import io
 from pathlib import Path
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 from PIL import Image

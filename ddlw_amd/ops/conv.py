@@ -9,8 +9,6 @@ table), ``hip`` (force hand-written kernels), ``stock`` (force MIOpen).
 from __future__ import annotations
 
 import os
-from typing import Optional
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

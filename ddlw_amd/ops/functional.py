@@ -1,12 +1,11 @@
-"""Python-facing fused ops, HIP-backed on GPU with CPU (stock-op) oracles.
+"""Functional index of the HIP-backed ops (see the modules that own them):
 
-Populated kernel by kernel; each function documents the reference op it
-replaces (SURVEY.md §2.4 kernel table) and has a parity test in
-``tests/test_kernels.py`` comparing the HIP path against the fp32 stock op.
+- fused BN(+add)+ReLU, maxpool, GAP, softmax-CE, u8 normalize: ``ops.layers``
+- conv fwd/dgrad/wgrad implicit-GEMM + depthwise: ``ops.conv`` / ``ops.conv_gemm``
+- fused SGD / Adam: ``ops.optim``
+- raw tensor-level wrappers: ``ops.binding``
 """
-from __future__ import annotations
-
-import torch
-
-# Kernel-backed autograd functions are registered here as the HIP suite lands
-# (bn_relu, maxpool, softmax_ce, fused_sgd, gap, conv2d implicit-GEMM).
+from .layers import (  # noqa: F401
+    normalize_u8_bf16,
+    softmax_cross_entropy,
+)

@@ -17,7 +17,6 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from . import binding
-from .runtime import has_lib
 
 
 def _hip_ops_disabled() -> bool:
