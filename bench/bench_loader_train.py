@@ -103,8 +103,10 @@ def main() -> None:
             step_fn()
         torch.cuda.synchronize()
         elapsed = time.perf_counter() - t0
+        it.close()  # stop the producer thread before teardown
 
     conv.delete()
+    torch.cuda.synchronize()
     out = {
         "metric": "images/sec (1 GPU) ResNet-50 224px, Parquet-loader-fed",
         "value": round(bs * steps / elapsed, 2),

@@ -197,11 +197,21 @@ class ShardedParquetLoader:
         def producer():
             try:
                 for batch in self._batches_cpu():
+                    # bounded put so the thread can exit promptly once the
+                    # consumer is gone (avoids a blocked thread at teardown)
+                    while not stop.is_set():
+                        try:
+                            q.put(batch, timeout=0.25)
+                            break
+                        except queue.Full:
+                            continue
                     if stop.is_set():
                         return
-                    q.put(batch)
             finally:
-                q.put(None)
+                try:
+                    q.put_nowait(None)
+                except queue.Full:
+                    pass
 
         t = threading.Thread(target=producer, daemon=True)
         t.start()
@@ -215,12 +225,14 @@ class ShardedParquetLoader:
                 yield di, dl
         finally:
             stop.set()
-            # drain so the producer can exit
+            # drain so the producer can exit, then join it so no thread is
+            # left inside native code at interpreter teardown
             while not q.empty():
                 try:
                     q.get_nowait()
                 except queue.Empty:
                     break
+            t.join(timeout=5.0)
 
 
 class Converter:
