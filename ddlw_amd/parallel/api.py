@@ -34,6 +34,7 @@ _initialized = False
 _world_size = 1
 _rank = 0
 _local_rank = 0
+_native_comm = None  # NativeComm when DDLW_NATIVE_CC=1 (see parallel.native)
 
 
 def init(backend: Optional[str] = None, timeout_s: float = 300.0) -> None:
@@ -49,8 +50,10 @@ def init(backend: Optional[str] = None, timeout_s: float = 300.0) -> None:
     _world_size = world
     if world > 1:
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
-        if backend == "nccl":
+            backend = os.environ.get("DDLW_BACKEND") or (
+                "nccl" if torch.cuda.is_available() else "gloo"
+            )
+        if torch.cuda.is_available():
             torch.cuda.set_device(_local_rank % torch.cuda.device_count())
         import datetime
 
@@ -58,11 +61,24 @@ def init(backend: Optional[str] = None, timeout_s: float = 300.0) -> None:
             backend=backend,
             timeout=datetime.timedelta(seconds=timeout_s),
         )
+        if backend == "nccl":
+            from . import native as _native_mod
+
+            if _native_mod.enabled():
+                # opt-in native RCCL data plane (DDLW_NATIVE_CC=1): gradient
+                # buckets + broadcasts go through libddlw_rccl on a side
+                # stream instead of ProcessGroupNCCL (both ARE RCCL/xGMI)
+                global _native_comm
+                device = torch.device("cuda", torch.cuda.current_device())
+                _native_comm = _native_mod.NativeComm(world, _rank, device)
     _initialized = True
 
 
 def shutdown() -> None:
-    global _initialized, _world_size, _rank, _local_rank
+    global _initialized, _world_size, _rank, _local_rank, _native_comm
+    if _native_comm is not None:
+        _native_comm.destroy()
+        _native_comm = None
     if _initialized and dist.is_initialized():
         dist.destroy_process_group()
     _initialized = False
@@ -129,7 +145,12 @@ def broadcast_parameters(module_or_params, root_rank: int = 0) -> None:
         tensors = list(module_or_params)
     for t in tensors:
         if t.dtype.is_floating_point or t.dtype in (torch.int64, torch.int32, torch.uint8):
-            dist.broadcast(t, src=root_rank)
+            if _native_comm is not None and t.is_cuda and t.is_contiguous():
+                _native_comm.broadcast_(t, root=root_rank)
+            else:
+                dist.broadcast(t, src=root_rank)
+    if _native_comm is not None:
+        torch.cuda.synchronize()
 
 
 def broadcast_optimizer_state(optimizer: torch.optim.Optimizer, root_rank: int = 0) -> None:
@@ -149,13 +170,14 @@ def broadcast_optimizer_state(optimizer: torch.optim.Optimizer, root_rank: int =
 
 
 class _Bucket:
-    __slots__ = ("params", "bytes", "flat", "work", "ready")
+    __slots__ = ("params", "bytes", "flat", "work", "ev", "ready")
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
         self.bytes = 0
         self.flat: Optional[torch.Tensor] = None
         self.work = None
+        self.ev = None  # native-path completion event
         self.ready = 0
 
 
@@ -230,9 +252,12 @@ class DistributedOptimizer:
                 f"allreduce[{flat.numel() * flat.element_size() >> 20}MiB]",
                 "collective", _t.time() * 1e6, 1.0, tid=_rank,
             )
-        # async: NCCL enqueues on the comm stream (overlaps backward); keep
-        # the work handle — wait() orders the current stream behind it
-        bucket.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
+        # async: RCCL enqueues on a comm stream (overlaps backward); the
+        # handle/event orders the optimizer stream behind the collective
+        if _native_comm is not None:
+            bucket.ev = _native_comm.allreduce_(flat, average=self.average)
+        else:
+            bucket.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
         bucket.flat = flat
 
     # -- torch optimizer surface ---------------------------------------- #
@@ -270,7 +295,11 @@ class DistributedOptimizer:
             if b.work is not None:
                 b.work.wait()
                 b.work = None
-            if self.average:
+            if b.ev is not None:
+                # native path: ncclAvg already averaged; just order streams
+                torch.cuda.current_stream().wait_event(b.ev)
+                b.ev = None
+            elif self.average:
                 b.flat.div_(_world_size)
             grads = [p.grad for p in b.params]
             for p, g in zip(
