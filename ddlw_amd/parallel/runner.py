@@ -57,9 +57,22 @@ class Runner:
     np >= 1 : spawn np local worker processes, return rank-0's result.
     """
 
-    def __init__(self, np: int, driver_log_verbosity: str = "all", timeout_s: float = 3600.0):
+    def __init__(
+        self,
+        np: int,
+        driver_log_verbosity: str = "all",
+        timeout_s: float = 3600.0,
+        max_restarts: int = 0,
+    ):
         self.np = np
         self.timeout_s = timeout_s
+        # SURVEY.md §5.3: the reference delegates fault tolerance to Spark
+        # barrier mode = gang-scheduled all-or-nothing job retry. Equivalent
+        # here: on any worker failure the gang is aborted and, with
+        # max_restarts > 0, relaunched whole; the train fn resumes from its
+        # last checkpoint (ModelCheckpoint + the broadcast callback restore
+        # rank parity, §5.4). Workers see DDLW_RESTART_ATTEMPT in env.
+        self.max_restarts = max_restarts
 
     def run(self, fn: Callable[..., Any], **kwargs) -> Any:
         if self.np == -1:
@@ -69,7 +82,23 @@ class Runner:
             return fn(**kwargs)
         if self.np < 1:
             raise ValueError(f"np must be -1 or >= 1, got {self.np}")
+        last_exc: Optional[BaseException] = None
+        for attempt in range(1 + self.max_restarts):
+            try:
+                return self._run_once(fn, kwargs, attempt)
+            except (RuntimeError, TimeoutError) as e:
+                last_exc = e
+                if attempt < self.max_restarts:
+                    print(
+                        f"[ddlw.Runner] job attempt {attempt} failed ({e}); "
+                        f"restarting gang ({self.max_restarts - attempt} retries left)",
+                        file=sys.stderr,
+                        flush=True,
+                    )
+        assert last_exc is not None
+        raise last_exc
 
+    def _run_once(self, fn: Callable[..., Any], kwargs: dict, attempt: int) -> Any:
         ctx = mp.get_context("spawn")
         result_q = ctx.SimpleQueue()
         port = _free_port()
@@ -81,6 +110,7 @@ class Runner:
                 "WORLD_SIZE": str(self.np),
                 "MASTER_ADDR": "127.0.0.1",
                 "MASTER_PORT": str(port),
+                "DDLW_RESTART_ATTEMPT": str(attempt),
             }
             p = ctx.Process(target=_worker_entry, args=(fn, kwargs, env, result_q), daemon=False)
             p.start()

@@ -137,3 +137,39 @@ def test_distributed_fused_sgd_matches_oracle():
         opt.step()
     for k, v in m.state_dict().items():
         assert torch.allclose(out[k], v, atol=1e-5), k
+
+
+def _elastic_worker(ckpt_dir):
+    """Crashes rank 1 on the first gang attempt after 'epoch 0' completes;
+    on restart, resumes from the checkpoint file (SURVEY.md §5.3 epoch-
+    granular restart)."""
+    attempt = int(os.environ["DDLW_RESTART_ATTEMPT"])
+    ckpt = os.path.join(ckpt_dir, "epoch.txt")
+    start_epoch = int(open(ckpt).read()) + 1 if os.path.exists(ckpt) else 0
+    done = start_epoch
+    for epoch in range(start_epoch, 3):
+        api.barrier()
+        if api.rank() == 0:  # rank-0-only checkpoint write (reference layout)
+            with open(ckpt, "w") as f:
+                f.write(str(epoch))
+        api.barrier()
+        done = epoch + 1
+        if attempt == 0 and api.rank() == 1 and epoch == 0:
+            os._exit(17)  # simulated hard worker death mid-job
+    return {"attempt": attempt, "resumed_from": start_epoch, "epochs_done": done}
+
+
+def test_runner_gang_restart_resumes_from_checkpoint(tmp_path):
+    out = Runner(np=2, timeout_s=120, max_restarts=1).run(
+        _elastic_worker, ckpt_dir=str(tmp_path)
+    )
+    assert out["attempt"] == 1          # first gang died, second succeeded
+    assert out["resumed_from"] == 1     # epoch 0's checkpoint was picked up
+    assert out["epochs_done"] == 3
+
+
+def test_runner_no_restart_still_aborts(tmp_path):
+    with pytest.raises(RuntimeError, match="exited with code"):
+        Runner(np=2, timeout_s=60, max_restarts=0).run(
+            _elastic_worker, ckpt_dir=str(tmp_path)
+        )
