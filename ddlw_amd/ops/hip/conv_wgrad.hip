@@ -1,5 +1,10 @@
 // MFMA implicit-GEMM weight gradient (wrw) for CDNA4 (gfx950), NHWC bf16.
 //
+// Reference parity: the conv weight gradients TF computes behind model.fit
+// for the trainable convs (SURVEY.md §2.4 K1-K3 "fwd+bwd for ResNet-50
+// BASELINE configs"; the reference itself holds its base frozen,
+// Part 1 .../02_model_training_single_node.py:167-169).
+//
 // dW[k][r][s][c] = sum_m dy[m][k] * x[m(r,s)][c]   (m = n*Ho*Wo rows)
 //
 // One GEMM per (r,s): dW_rs[K][C] = dy^T @ x_shifted, reduction over m.
