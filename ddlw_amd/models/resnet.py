@@ -40,6 +40,12 @@ class Bottleneck(nn.Module):
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops import block as _block
+
+        if _block.bottleneck_fusable(self, x):
+            # whole-block fused Function: the residual-join gradient add is
+            # absorbed into conv1's dgrad epilogue (see ops/block.py)
+            return _block.bottleneck_forward(self, x)
         identity = self.downsample(x) if self.downsample is not None else x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))

@@ -58,7 +58,12 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     int R, int S, int stride, int pad, int nwg_swz,
     int oH, int oW, int oS,  // output scatter: flat out row = (n*oH + ho*oS)*oW + wo*oS
     unsigned long long magic_wo, unsigned shift_wo,
-    unsigned long long magic_ho, unsigned shift_ho) {
+    unsigned long long magic_ho, unsigned shift_ho,
+    const bf16_t* __restrict__ accp) {  // optional epilogue accumulate input
+  // accp != nullptr: y = conv + accp (read at the output index). Used to
+  // fuse the residual-join gradient add (d_block_input = conv1_dgrad +
+  // d_identity) into the dgrad epilogue — saves the engine's separate
+  // 3-pass elementwise add (read A, read B, write C) per ResNet join.
   constexpr int BK = 64;
   constexpr int WM = BM / 2, WN = BN / 2;
   constexpr int MF = WM / 16, NF = WN / 16;
@@ -280,6 +285,11 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   // 32x32 D map: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
   const int d_col32 = lane & 31;
   const int d_rbase32 = 4 * (lane >> 5);
+  auto b2f = [](bf16_t u) -> float {
+    union { unsigned i; float f; } cvt;
+    cvt.i = (unsigned)u << 16;
+    return cvt.f;
+  };
   if (!EPI_LDS) {
     if constexpr (M32) {
       #pragma unroll
@@ -292,7 +302,12 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           for (int reg = 0; reg < 16; ++reg) {
             int row = (reg & 3) + 8 * (reg >> 2) + d_rbase32;
             long m = tile_m * BM + wr * WM + mi * 32 + row;
-            if (m < M) y[out_row(m) * K + j] = f2b_hw(acc2[mi][ni][reg]);
+            if (m < M) {
+              long oi = out_row(m) * K + j;
+              float v = acc2[mi][ni][reg];
+              if (accp) v += b2f(accp[oi]);
+              y[oi] = f2b_hw(v);
+            }
           }
         }
       }
@@ -306,7 +321,12 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           #pragma unroll
           for (int q = 0; q < 4; ++q) {
             long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
-            if (m < M) y[out_row(m) * K + j] = f2b_hw(acc[mi][ni][q]);
+            if (m < M) {
+              long oi = out_row(m) * K + j;
+              float v = acc[mi][ni][q];
+              if (accp) v += b2f(accp[oi]);
+              y[oi] = f2b_hw(v);
+            }
           }
         }
       }
@@ -351,6 +371,21 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     if (m < M) {
       const long orow = out_row(m);
       if (j_base + 8 <= K) {
+        if (accp) {
+          // accumulate: coalesced uint4 read of the add-input at the same
+          // address, pairwise bf16 add in fp32 (matches the engine's
+          // bf16+bf16 add numerics)
+          uint4 a = *reinterpret_cast<const uint4*>(accp + orow * K + j_base);
+          auto addpair = [&](unsigned v, unsigned aw) -> unsigned {
+            bf16_t lo = f2b_hw(b2f((bf16_t)(v & 0xffff)) + b2f((bf16_t)(aw & 0xffff)));
+            bf16_t hi = f2b_hw(b2f((bf16_t)(v >> 16)) + b2f((bf16_t)(aw >> 16)));
+            return (unsigned)lo | ((unsigned)hi << 16);
+          };
+          val.x = addpair(val.x, a.x);
+          val.y = addpair(val.y, a.y);
+          val.z = addpair(val.z, a.z);
+          val.w = addpair(val.w, a.w);
+        }
         *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
       } else {
         // static component extraction (a reinterpret pointer into `val`
@@ -359,7 +394,10 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
         for (int e = 0; e < 8; ++e) {
           if (j_base + e < K) {
             unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
-            y[orow * K + j_base + e] = (bf16_t)(wd >> ((e & 1) * 16));
+            bf16_t ov = (bf16_t)(wd >> ((e & 1) * 16));
+            if (accp)
+              ov = f2b_hw(b2f(ov) + b2f(accp[orow * K + j_base + e]));
+            y[orow * K + j_base + e] = ov;
           }
         }
       }
@@ -382,12 +420,12 @@ static inline void make_magic(unsigned d, unsigned long long* magic, unsigned* s
   *shift = 32 + s;
 }
 
-DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
-                                    const void* zpage,
-                                    int N, int H, int W_, int C, int K,
-                                    int Ho, int Wo, int R, int S, int stride,
-                                    int pad, int oH, int oW, int oS,
-                                    void* stream) {
+DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
+                                        const void* zpage, const void* acc,
+                                        int N, int H, int W_, int C, int K,
+                                        int Ho, int Wo, int R, int S, int stride,
+                                        int pad, int oH, int oW, int oS,
+                                        void* stream) {
   if (C % 64 != 0) {
     ddlw_set_error("conv_fwd_igemm: C must be a multiple of 64");
     return 2;
@@ -420,13 +458,15 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
                          dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
                          (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
                          N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
-                         (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho);  \
+                         (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho,   \
+                         (const bf16_t*)acc);                                 \
     else                                                                      \
       hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, true>),         \
                          dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
                          (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
                          N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
-                         (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho);  \
+                         (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho,   \
+                         (const bf16_t*)acc);                                 \
   } while (0)
   if (K >= 128) {
     if (T == 1) LAUNCH(128, 128, true, 1);
@@ -443,4 +483,14 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
   }
 #undef LAUNCH
   DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
+                                    const void* zpage,
+                                    int N, int H, int W_, int C, int K,
+                                    int Ho, int Wo, int R, int S, int stride,
+                                    int pad, int oH, int oW, int oS,
+                                    void* stream) {
+  return ddlw_conv_fwd_igemm_acc(x, w, y, zpage, nullptr, N, H, W_, C, K, Ho,
+                                 Wo, R, S, stride, pad, oH, oW, oS, stream);
 }

@@ -233,10 +233,14 @@ def softmax_cross_entropy(logits: torch.Tensor, labels: torch.Tensor) -> torch.T
 def normalize_u8_bf16(x_u8: torch.Tensor) -> torch.Tensor:
     """uint8 NHWC (channels_last) -> bf16 NHWC, x/127.5-1 in one pass."""
     if x_u8.is_cuda and not _hip_ops_disabled() and x_u8.numel() % 16 == 0:
-        out = torch.empty(x_u8.shape, dtype=torch.bfloat16, device=x_u8.device)
         if x_u8.dim() == 4:
-            out = out.to(memory_format=torch.channels_last)
             assert x_u8.is_contiguous(memory_format=torch.channels_last)
+            out = torch.empty(
+                x_u8.shape, dtype=torch.bfloat16, device=x_u8.device,
+                memory_format=torch.channels_last,
+            )
+        else:
+            out = torch.empty(x_u8.shape, dtype=torch.bfloat16, device=x_u8.device)
         binding.normalize_u8(x_u8, out)
         return out
     return (x_u8.to(torch.bfloat16) / 127.5) - 1.0

@@ -159,3 +159,98 @@ def test_mobilenet_base_uses_depthwise_kernel():
         y = mb(x32.to(torch.bfloat16).contiguous(memory_format=torch.channels_last))
     rel = (y.float() - ref).abs().max() / (ref.abs().max() + 1e-6)
     assert rel.item() < 0.12, rel.item()
+
+
+def test_conv_dgrad_acc_fusion_parity():
+    """Epilogue accumulate: dgrad(dy, w) + acc in one kernel pass."""
+    from ddlw_amd.ops import conv_gemm
+
+    torch.manual_seed(5)
+    for C, K, T_long in ((64, 256, False), (512, 256, True)):
+        w = _cl(torch.randn(K, C, 1, 1, device=_cuda()).to(torch.bfloat16))
+        dy = _cl(torch.randn(4, K, 14, 14, device=_cuda()).to(torch.bfloat16))
+        acc = _cl(torch.randn(4, C, 14, 14, device=_cuda()).to(torch.bfloat16))
+        fused = conv_gemm.conv_dgrad_kernel(dy, w, (4, C, 14, 14), 0, 1, acc=acc)
+        plain = conv_gemm.conv_dgrad_kernel(dy, w, (4, C, 14, 14), 0, 1)
+        ref = (plain.float() + acc.float())
+        scale = ref.abs().max() + 1e-6
+        assert ((fused.float() - ref).abs().max() / scale).item() < 2e-2
+
+
+def _make_block(in_ch=256, mid=64, seed=9):
+    from ddlw_amd.models.resnet import Bottleneck
+
+    torch.manual_seed(seed)
+    blk = Bottleneck(in_ch, mid)
+    # non-trivial BN state so the fused stats path is exercised
+    for bn in (blk.bn1, blk.bn2, blk.bn3):
+        torch.nn.init.uniform_(bn.weight, 0.5, 1.5)
+        torch.nn.init.uniform_(bn.bias, -0.2, 0.2)
+    return blk
+
+
+@pytest.mark.parametrize("conv_mode", ["hip", "auto"])
+def test_fused_bottleneck_matches_unfused(monkeypatch, conv_mode):
+    """Whole-block fused Function vs the per-layer HIP path: same kernels,
+    same results (join-add rounding is the only difference)."""
+    import copy
+
+    monkeypatch.setenv("DDLW_CONV", conv_mode)
+    blk = _make_block().to(_cuda()).to(memory_format=torch.channels_last)
+    for m in blk.modules():
+        if isinstance(m, torch.nn.Conv2d):
+            m.to(torch.bfloat16)
+    blk2 = copy.deepcopy(blk)
+    x = _cl(torch.randn(4, 256, 28, 28, device=_cuda()).to(torch.bfloat16))
+
+    monkeypatch.setenv("DDLW_FUSED_BLOCK", "1")
+    blk.train()
+    xa = x.clone().requires_grad_(True)
+    out_f = blk(xa)
+    out_f.float().square().mean().backward()
+
+    monkeypatch.setenv("DDLW_FUSED_BLOCK", "0")
+    blk2.train()
+    xb = x.clone().requires_grad_(True)
+    out_u = blk2(xb)
+    out_u.float().square().mean().backward()
+
+    assert torch.equal(out_f, out_u)  # forward is the identical kernel chain
+    for (n1, p1), (_, p2) in zip(blk.named_parameters(), blk2.named_parameters()):
+        s = p2.grad.float().abs().max() + 1e-6
+        assert ((p1.grad.float() - p2.grad.float()).abs().max() / s).item() < 2e-2, n1
+    s = xb.grad.float().abs().max() + 1e-6
+    assert ((xa.grad.float() - xb.grad.float()).abs().max() / s).item() < 2e-2
+    # running stats updated identically
+    assert torch.allclose(blk.bn1.running_mean, blk2.bn1.running_mean)
+    assert int(blk.bn3.num_batches_tracked) == int(blk2.bn3.num_batches_tracked)
+
+
+def test_fused_bottleneck_vs_fp32_oracle(monkeypatch):
+    """Fused block vs the stock fp32 CPU-oracle path on the same weights."""
+    import copy
+
+    monkeypatch.setenv("DDLW_FUSED_BLOCK", "1")
+    blk = _make_block(seed=11)
+    oracle = copy.deepcopy(blk).float()
+    blk = blk.to(_cuda()).to(memory_format=torch.channels_last)
+    for m in blk.modules():
+        if isinstance(m, torch.nn.Conv2d):
+            m.to(torch.bfloat16)
+    x32 = torch.randn(4, 256, 28, 28)
+    x = _cl(x32.to(_cuda()).to(torch.bfloat16))
+
+    blk.train()
+    xa = x.clone().requires_grad_(True)
+    out = blk(xa)
+    out.float().mean().backward()
+
+    oracle.train()
+    xo = x.float().cpu().requires_grad_(True)  # same bf16-rounded input
+    out_o = oracle(xo)
+    out_o.mean().backward()
+
+    s = out_o.abs().max().item() + 1e-3
+    assert ((out.float().cpu() - out_o).abs().max() / s).item() < 6e-2
+    so = xo.grad.abs().max().item() + 1e-6
+    assert ((xa.grad.float().cpu() - xo.grad).abs().max() / so).item() < 8e-2
