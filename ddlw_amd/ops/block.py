@@ -1,15 +1,17 @@
-"""Fused identity-bottleneck block: one autograd Function for the whole
-conv1-bn1-relu / conv2-bn2-relu / conv3-bn3(+res)+relu residual block.
+"""Fused bottleneck block: one autograd Function for the whole
+conv1-bn1-relu / conv2-bn2-relu / conv3-bn3(+res)+relu residual block
+(identity shortcut or conv+BN downsample shortcut).
 
 Why this exists (MI355X-first): with per-layer autograd Functions, the
 gradient w.r.t. the block input arrives from TWO paths — conv1's dgrad and
-the residual pass-through (``dres`` from the fused bn3 backward) — and the
-autograd engine materialises both then sums them with an elementwise add
-(3 HBM passes over an N*C*H*W tensor per join; ~4% of the ResNet-50 step
-in the steady-state profile). Owning the whole block's backward lets the
-join-add ride conv1's dgrad epilogue (``conv_backward(..., acc=dres)``):
-the add becomes one extra coalesced read in a kernel that was writing that
-tensor anyway.
+the shortcut path (``dres`` from the fused bn3 backward, possibly through
+the downsample conv's dgrad) — and the autograd engine materialises both
+then sums them with an elementwise add (3 HBM passes over an N*C*H*W tensor
+per join; ~4% of the ResNet-50 step in the steady-state profile). Owning
+the whole block's backward lets the join-add ride conv1's dgrad epilogue
+(``conv_backward(..., acc=...)``): the add becomes one extra coalesced read
+in a kernel that was writing that tensor anyway. Measured +1% whole-model
+step throughput on 1x MI355X.
 
 Numerics match the per-layer path: identical kernels in the identical
 order; the only difference is where the (bf16 + bf16) join-add rounds, and
@@ -17,11 +19,10 @@ the fused path rounds from the fp32 accumulator (>= the engine-add's
 precision). Covered by ``tests/test_conv_kernels.py`` parity tests vs both
 the stock fp32 oracle and the unfused HIP path.
 
-Enabled by default on the HIP training path for identity-shortcut blocks;
-``DDLW_FUSED_BLOCK=0`` disables (falls back to per-layer Functions).
-Reference scope: this fuses the backward of the reference's K1-K5 kernel
-chain at block granularity; the reference itself leaves
-fusion to XLA.
+Enabled by default on the HIP training path; ``DDLW_FUSED_BLOCK=0``
+disables (falls back to per-layer Functions). Reference scope: this fuses
+the backward of the reference's K1-K5 kernel chain at
+block granularity; the reference itself leaves fusion to XLA.
 """
 from __future__ import annotations
 
@@ -47,13 +48,30 @@ def _fwd_conv(x, w, stride, padding, hip_fwd):
 
 
 class _BottleneckFn(torch.autograd.Function):
-    """Training-mode fused bottleneck (identity shortcut, stride 1)."""
+    """Training-mode fused bottleneck. ``wd/gd/bd`` are the downsample
+    conv/BN parameters (None for identity-shortcut blocks)."""
 
     @staticmethod
-    def forward(ctx, x, w1, g1, b1, w2, g2, b2, w3, g3, b3, block):
+    def forward(ctx, x, w1, g1, b1, w2, g2, b2, w3, g3, b3, wd, gd, bd, block):
         bn1, bn2, bn3 = block.bn1, block.bn2, block.bn3
+        stride = block.stride
         mode = os.environ.get("DDLW_CONV", "auto")
         x = _cl(x)
+
+        if wd is not None:
+            dsc = block.downsample.conv
+            dsbn = block.downsample.bn
+            conv_gemm.available(dsc, x, mode)
+            rdf, rdd, rdg = dsc._ddlw_route
+            td = _fwd_conv(x, wd, stride, 0, rdf)
+            md, sd = binding.bn_stats(td, dsbn.eps, dsbn.momentum,
+                                      dsbn.running_mean, dsbn.running_var)
+            res, _ = binding.bn_apply(td, None, md, sd, gd, bd, False)
+            dsbn.num_batches_tracked += 1
+        else:
+            rdd = rdg = False
+            td = md = sd = None
+            res = x
 
         conv_gemm.available(block.conv1, x, mode)
         r1f, r1d, r1g = block.conv1._ddlw_route
@@ -64,7 +82,7 @@ class _BottleneckFn(torch.autograd.Function):
 
         conv_gemm.available(block.conv2, a1, mode)
         r2f, r2d, r2g = block.conv2._ddlw_route
-        t2 = _fwd_conv(a1, w2, 1, 1, r2f)
+        t2 = _fwd_conv(a1, w2, stride, 1, r2f)
         m2, s2 = binding.bn_stats(t2, bn2.eps, bn2.momentum,
                                   bn2.running_mean, bn2.running_var)
         a2, mask2 = binding.bn_apply(t2, None, m2, s2, g2, b2, True)
@@ -74,29 +92,31 @@ class _BottleneckFn(torch.autograd.Function):
         t3 = _fwd_conv(a2, w3, 1, 0, r3f)
         m3, s3 = binding.bn_stats(t3, bn3.eps, bn3.momentum,
                                   bn3.running_mean, bn3.running_var)
-        out, mask3 = binding.bn_apply(t3, x, m3, s3, g3, b3, True)
+        out, mask3 = binding.bn_apply(t3, res, m3, s3, g3, b3, True)
 
         for bn in (bn1, bn2, bn3):
             bn.num_batches_tracked += 1
-        ctx.save_for_backward(x, w1, g1, w2, g2, w3, g3,
-                              t1, a1, t2, a2, t3,
+        ctx.save_for_backward(x, w1, g1, w2, g2, w3, g3, wd, gd,
+                              t1, a1, t2, a2, t3, td,
                               mask1, mask2, mask3,
-                              m1, s1, m2, s2, m3, s3)
-        ctx.routes = ((r1d, r1g), (r2d, r2g), (r3d, r3g))
+                              m1, s1, m2, s2, m3, s3, md, sd)
+        ctx.routes = ((r1d, r1g), (r2d, r2g), (r3d, r3g), (rdd, rdg))
+        ctx.stride = stride
         return out
 
     @staticmethod
     def backward(ctx, dy):
-        (x, w1, g1, w2, g2, w3, g3,
-         t1, a1, t2, a2, t3,
+        (x, w1, g1, w2, g2, w3, g3, wd, gd,
+         t1, a1, t2, a2, t3, td,
          mask1, mask2, mask3,
-         m1, s1, m2, s2, m3, s3) = ctx.saved_tensors
-        (r1d, r1g), (r2d, r2g), (r3d, r3g) = ctx.routes
+         m1, s1, m2, s2, m3, s3, md, sd) = ctx.saved_tensors
+        (r1d, r1g), (r2d, r2g), (r3d, r3g), (rdd, rdg) = ctx.routes
+        stride = ctx.stride
         dy = _cl(dy)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
 
-        # bn3 (+residual, +relu): dres is the pass-through gradient
+        # bn3 (+residual, +relu): dres is the shortcut-path gradient
         db3, dg3 = binding.bn_bwd_reduce(dy, mask3, t3, m3, s3, True)
         dt3, dres = binding.bn_bwd_dx(dy, mask3, t3, m3, s3, g3, db3, dg3,
                                       True, True)
@@ -105,21 +125,33 @@ class _BottleneckFn(torch.autograd.Function):
         db2, dg2 = binding.bn_bwd_reduce(da2, mask2, t2, m2, s2, True)
         dt2, _ = binding.bn_bwd_dx(da2, mask2, t2, m2, s2, g2, db2, dg2,
                                    True, False)
-        da1, dw2 = conv_gemm.conv_backward(dt2, a1, w2, 1, 1, r2d, r2g)
+        da1, dw2 = conv_gemm.conv_backward(dt2, a1, w2, stride, 1, r2d, r2g)
 
         db1, dg1 = binding.bn_bwd_reduce(da1, mask1, t1, m1, s1, True)
         dt1, _ = binding.bn_bwd_dx(da1, mask1, t1, m1, s1, g1, db1, dg1,
                                    True, False)
-        # the join-add rides conv1's dgrad epilogue (acc=dres)
-        dx, dw1 = conv_gemm.conv_backward(dt1, x, w1, 1, 0, r1d, r1g, acc=dres)
 
-        return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3, None)
+        dwd = dgd = dbd = None
+        if wd is not None:
+            # downsample shortcut: dres -> bn_d bwd -> conv_d dgrad -> join
+            dbd, dgd = binding.bn_bwd_reduce(dres, None, td, md, sd, False)
+            dtd, _ = binding.bn_bwd_dx(dres, None, td, md, sd, gd, dbd, dgd,
+                                       False, False)
+            dxd, dwd = conv_gemm.conv_backward(dtd, x, wd, stride, 0, rdd, rdg)
+            join = dxd
+        else:
+            join = dres
+        # the join-add rides conv1's dgrad epilogue (acc=join)
+        dx, dw1 = conv_gemm.conv_backward(dt1, x, w1, 1, 0, r1d, r1g, acc=join)
+
+        return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
+                dwd, dgd, dbd, None)
 
 
 def bottleneck_fusable(block, x: torch.Tensor) -> bool:
-    """Fused path: training, grad-enabled, identity shortcut, bf16 CUDA
-    input, fp32 BN params/stats, kernel-supported channel counts."""
-    if not fused_block_enabled() or block.downsample is not None:
+    """Fused path: training, grad-enabled, bf16 CUDA input, fp32 BN
+    params/stats, kernel-supported channel counts."""
+    if not fused_block_enabled():
         return False
     if not (block.training and torch.is_grad_enabled()):
         return False
@@ -127,7 +159,10 @@ def bottleneck_fusable(block, x: torch.Tensor) -> bool:
         return False
     if os.environ.get("DDLW_DISABLE_HIP_OPS", "0") == "1":
         return False
-    for bn in (block.bn1, block.bn2, block.bn3):
+    bns = [block.bn1, block.bn2, block.bn3]
+    if block.downsample is not None:
+        bns.append(block.downsample.bn)
+    for bn in bns:
         if bn.weight.dtype != torch.float32 or bn.running_mean.dtype != torch.float32:
             return False
         if not binding.supported_channels(bn.num_features):
@@ -136,10 +171,17 @@ def bottleneck_fusable(block, x: torch.Tensor) -> bool:
 
 
 def bottleneck_forward(block, x: torch.Tensor) -> torch.Tensor:
+    if block.downsample is not None:
+        wd = block.downsample.conv.weight
+        gd = block.downsample.bn.weight
+        bd = block.downsample.bn.bias
+    else:
+        wd = gd = bd = None
     return _BottleneckFn.apply(
         x,
         block.conv1.weight, block.bn1.weight, block.bn1.bias,
         block.conv2.weight, block.bn2.weight, block.bn2.bias,
         block.conv3.weight, block.bn3.weight, block.bn3.bias,
+        wd, gd, bd,
         block,
     )

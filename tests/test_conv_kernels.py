@@ -177,26 +177,31 @@ def test_conv_dgrad_acc_fusion_parity():
         assert ((fused.float() - ref).abs().max() / scale).item() < 2e-2
 
 
-def _make_block(in_ch=256, mid=64, seed=9):
-    from ddlw_amd.models.resnet import Bottleneck
+def _make_block(in_ch=256, mid=64, seed=9, stride=1, downsample=False):
+    from ddlw_amd.models.resnet import Bottleneck, Downsample
 
     torch.manual_seed(seed)
-    blk = Bottleneck(in_ch, mid)
+    ds = Downsample(in_ch, mid * 4, stride) if downsample else None
+    blk = Bottleneck(in_ch, mid, stride=stride, downsample=ds)
     # non-trivial BN state so the fused stats path is exercised
-    for bn in (blk.bn1, blk.bn2, blk.bn3):
+    bns = [blk.bn1, blk.bn2, blk.bn3] + ([ds.bn] if ds else [])
+    for bn in bns:
         torch.nn.init.uniform_(bn.weight, 0.5, 1.5)
         torch.nn.init.uniform_(bn.bias, -0.2, 0.2)
     return blk
 
 
 @pytest.mark.parametrize("conv_mode", ["hip", "auto"])
-def test_fused_bottleneck_matches_unfused(monkeypatch, conv_mode):
+@pytest.mark.parametrize("downsample", [False, True])
+def test_fused_bottleneck_matches_unfused(monkeypatch, conv_mode, downsample):
     """Whole-block fused Function vs the per-layer HIP path: same kernels,
     same results (join-add rounding is the only difference)."""
     import copy
 
     monkeypatch.setenv("DDLW_CONV", conv_mode)
-    blk = _make_block().to(_cuda()).to(memory_format=torch.channels_last)
+    blk = _make_block(
+        stride=2 if downsample else 1, downsample=downsample
+    ).to(_cuda()).to(memory_format=torch.channels_last)
     for m in blk.modules():
         if isinstance(m, torch.nn.Conv2d):
             m.to(torch.bfloat16)
@@ -215,7 +220,14 @@ def test_fused_bottleneck_matches_unfused(monkeypatch, conv_mode):
     out_u = blk2(xb)
     out_u.float().square().mean().backward()
 
-    assert torch.equal(out_f, out_u)  # forward is the identical kernel chain
+    if conv_mode == "hip":
+        # identical deterministic ddlw kernel chain -> bit-equal
+        assert torch.equal(out_f, out_u)
+    else:
+        # auto may route convs to MIOpen, whose algo choice is not
+        # guaranteed bit-deterministic across calls
+        sf = out_u.float().abs().max() + 1e-6
+        assert ((out_f.float() - out_u.float()).abs().max() / sf).item() < 1e-2
     for (n1, p1), (_, p2) in zip(blk.named_parameters(), blk2.named_parameters()):
         s = p2.grad.float().abs().max() + 1e-6
         assert ((p1.grad.float() - p2.grad.float()).abs().max() / s).item() < 2e-2, n1
@@ -243,14 +255,16 @@ def test_fused_bottleneck_vs_fp32_oracle(monkeypatch):
     blk.train()
     xa = x.clone().requires_grad_(True)
     out = blk(xa)
-    out.float().mean().backward()
+    out.float().square().mean().backward()
 
     oracle.train()
     xo = x.float().cpu().requires_grad_(True)  # same bf16-rounded input
     out_o = oracle(xo)
-    out_o.mean().backward()
+    out_o.square().mean().backward()
+    # note: the two losses differ by the bf16 rounding of `out`, so grads
+    # carry that perturbation on top of kernel rounding — tolerance reflects it
 
     s = out_o.abs().max().item() + 1e-3
     assert ((out.float().cpu() - out_o).abs().max() / s).item() < 6e-2
     so = xo.grad.abs().max().item() + 1e-6
-    assert ((xa.grad.float().cpu() - xo.grad).abs().max() / so).item() < 8e-2
+    assert ((xa.grad.float().cpu() - xo.grad).abs().max() / so).item() < 1.5e-1
