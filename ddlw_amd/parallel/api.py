@@ -143,10 +143,17 @@ def broadcast_parameters(module_or_params, root_rank: int = 0) -> None:
         tensors: Iterable[torch.Tensor] = list(module_or_params.state_dict().values())
     else:
         tensors = list(module_or_params)
+    use_nccl = dist.get_backend() == "nccl"
     for t in tensors:
         if t.dtype.is_floating_point or t.dtype in (torch.int64, torch.int32, torch.uint8):
             if _native_comm is not None and t.is_cuda and t.is_contiguous():
                 _native_comm.broadcast_(t, root=root_rank)
+            elif use_nccl and not t.is_cuda:
+                # CPU-resident buffers (e.g. BN step counters) ride a device
+                # round-trip — NCCL/RCCL moves device tensors only
+                d = t.to(torch.device("cuda", torch.cuda.current_device()))
+                dist.broadcast(d, src=root_rank)
+                t.copy_(d.cpu())
             else:
                 dist.broadcast(t, src=root_rank)
     if _native_comm is not None:
