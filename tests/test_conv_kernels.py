@@ -233,8 +233,9 @@ def test_fused_bottleneck_matches_unfused(monkeypatch, conv_mode, downsample):
         assert ((p1.grad.float() - p2.grad.float()).abs().max() / s).item() < 2e-2, n1
     s = xb.grad.float().abs().max() + 1e-6
     assert ((xa.grad.float() - xb.grad.float()).abs().max() / s).item() < 2e-2
-    # running stats updated identically
-    assert torch.allclose(blk.bn1.running_mean, blk2.bn1.running_mean)
+    # running stats updated identically (loose atol: under auto, MIOpen's
+    # algo choice perturbs the conv output in the last bf16 ulp)
+    assert torch.allclose(blk.bn1.running_mean, blk2.bn1.running_mean, atol=1e-4)
     assert int(blk.bn3.num_batches_tracked) == int(blk2.bn3.num_batches_tracked)
 
 
