@@ -62,7 +62,7 @@ class _BottleneckFn(torch.autograd.Function):
             dsc = block.downsample.conv
             dsbn = block.downsample.bn
             conv_gemm.available(dsc, x, mode)
-            rdf, rdd, rdg = dsc._ddlw_route
+            rdf, rdd, rdg = getattr(dsc, "_ddlw_route", (False, False, False))
             td = _fwd_conv(x, wd, stride, 0, rdf)
             md, sd = binding.bn_stats(td, dsbn.eps, dsbn.momentum,
                                       dsbn.running_mean, dsbn.running_var)
@@ -74,21 +74,21 @@ class _BottleneckFn(torch.autograd.Function):
             res = x
 
         conv_gemm.available(block.conv1, x, mode)
-        r1f, r1d, r1g = block.conv1._ddlw_route
+        r1f, r1d, r1g = getattr(block.conv1, "_ddlw_route", (False, False, False))
         t1 = _fwd_conv(x, w1, 1, 0, r1f)
         m1, s1 = binding.bn_stats(t1, bn1.eps, bn1.momentum,
                                   bn1.running_mean, bn1.running_var)
         a1, mask1 = binding.bn_apply(t1, None, m1, s1, g1, b1, True)
 
         conv_gemm.available(block.conv2, a1, mode)
-        r2f, r2d, r2g = block.conv2._ddlw_route
+        r2f, r2d, r2g = getattr(block.conv2, "_ddlw_route", (False, False, False))
         t2 = _fwd_conv(a1, w2, stride, 1, r2f)
         m2, s2 = binding.bn_stats(t2, bn2.eps, bn2.momentum,
                                   bn2.running_mean, bn2.running_var)
         a2, mask2 = binding.bn_apply(t2, None, m2, s2, g2, b2, True)
 
         conv_gemm.available(block.conv3, a2, mode)
-        r3f, r3d, r3g = block.conv3._ddlw_route
+        r3f, r3d, r3g = getattr(block.conv3, "_ddlw_route", (False, False, False))
         t3 = _fwd_conv(a2, w3, 1, 0, r3f)
         m3, s3 = binding.bn_stats(t3, bn3.eps, bn3.momentum,
                                   bn3.running_mean, bn3.running_var)
