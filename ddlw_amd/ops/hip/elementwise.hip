@@ -137,6 +137,11 @@ __global__ __launch_bounds__(256) void k_bn_finalize(
 //   y = act((x - mean[c]) * rstd[c] * gamma[c] + beta[c] [+ res])
 // RELU: 0 = identity, 1 = relu. res may be null.
 // ---------------------------------------------------------------------------
+// Geometry (shared with k_bn_bwd_dx): each thread owns ONE fixed 8-channel
+// vector column and streams rows, so the per-channel tables are loaded ONCE
+// into registers and folded (y = x*scale + shift) — the naive grid-stride
+// form reloads 4 tables x 8 scalars per 16-B vector (issue-bound, measured
+// 1.6x off the HBM roofline on the 56^2 layers).
 template <int RELU, bool HAS_RES>
 __global__ __launch_bounds__(256) void k_bn_apply(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ res,
@@ -145,18 +150,29 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     const float* __restrict__ gamma, const float* __restrict__ beta,
     long rows, int C) {
   const int vecC = C >> 3;
-  const long nvec = rows * vecC;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
-       i += (long)gridDim.x * blockDim.x) {
-    const int vc = (int)(i % vecC);
+  const int VPB = vecC < 256 ? vecC : 256;
+  const int ROWG = 256 / VPB;
+  const int tid = threadIdx.x;
+  const int vec = blockIdx.x * VPB + (tid % VPB);
+  const int rowg = tid / VPB;
+  if (rowg >= ROWG || vec >= vecC) return;
+  float scale[8], shift[8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    const int c = vec * 8 + k;
+    scale[k] = rstd[c] * gamma[c];
+    shift[k] = beta[c] - mean[c] * scale[k];
+  }
+  const long rstride = (long)gridDim.y * ROWG;
+  for (long r = (long)blockIdx.y * ROWG + rowg; r < rows; r += rstride) {
+    const long base = r * C + (long)vec * 8;
     bf16x8 v, o, rv;
-    v.v = *reinterpret_cast<const uint4*>(x + i * 8);
-    if (HAS_RES) rv.v = *reinterpret_cast<const uint4*>(res + i * 8);
+    v.v = *reinterpret_cast<const uint4*>(x + base);
+    if (HAS_RES) rv.v = *reinterpret_cast<const uint4*>(res + base);
     unsigned char mb = 0;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      const int c = vc * 8 + k;
-      float f = (b2f(v.h[k]) - mean[c]) * rstd[c] * gamma[c] + beta[c];
+      float f = b2f(v.h[k]) * scale[k] + shift[k];
       if (HAS_RES) f += b2f(rv.h[k]);
       if (RELU) {
         if (f > 0.f) mb |= (1u << k);
@@ -164,9 +180,9 @@ __global__ __launch_bounds__(256) void k_bn_apply(
       }
       o.h[k] = f2b(f);
     }
-    *reinterpret_cast<uint4*>(y + i * 8) = o.v;
-    if (RELU) mask[i] = mb;  // 1 byte per 8-channel vector: bwd re-reads
-                             // this instead of the whole y tensor
+    *reinterpret_cast<uint4*>(y + base) = o.v;
+    if (RELU) mask[r * vecC + vec] = mb;  // 1 byte per 8-channel vector:
+                                          // bwd reads this instead of y
   }
 }
 
@@ -277,6 +293,9 @@ __global__ __launch_bounds__(256) void k_bn_grad_finalize(
 //   dx   = gamma*rstd * (dy_m - dbeta/M - xhat * dgamma/M)
 //   dres = dy_m (residual branch gets the masked upstream grad)
 // ---------------------------------------------------------------------------
+// Same fixed-channel geometry as k_bn_apply; the per-channel terms fold to
+//   dx = A*dy_m - B*x + D   with A = gamma*rstd, B = A*rstd*dgamma/M,
+//   D = -A*dbeta/M + B*mean  (2 FMA per element, tables in registers).
 template <int RELU, bool HAS_RES>
 __global__ __launch_bounds__(256) void k_bn_bwd_dx(
     const bf16_t* __restrict__ dy, const unsigned char* __restrict__ mask,
@@ -287,27 +306,38 @@ __global__ __launch_bounds__(256) void k_bn_bwd_dx(
     bf16_t* __restrict__ dx, bf16_t* __restrict__ dres,
     long rows, int C) {
   const int vecC = C >> 3;
-  const long nvec = rows * vecC;
+  const int VPB = vecC < 256 ? vecC : 256;
+  const int ROWG = 256 / VPB;
+  const int tid = threadIdx.x;
+  const int vec = blockIdx.x * VPB + (tid % VPB);
+  const int rowg = tid / VPB;
+  if (rowg >= ROWG || vec >= vecC) return;
   const float invM = 1.f / (float)rows;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
-       i += (long)gridDim.x * blockDim.x) {
-    const int vc = (int)(i % vecC);
+  float A[8], B[8], D[8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    const int c = vec * 8 + k;
+    A[k] = gamma[c] * rstd[c];
+    B[k] = A[k] * rstd[c] * dgamma[c] * invM;
+    D[k] = B[k] * mean[c] - A[k] * dbeta[c] * invM;
+  }
+  const long rstride = (long)gridDim.y * ROWG;
+  for (long r = (long)blockIdx.y * ROWG + rowg; r < rows; r += rstride) {
+    const long base = r * C + (long)vec * 8;
     bf16x8 vdy, vx, odx, odr;
-    vdy.v = *reinterpret_cast<const uint4*>(dy + i * 8);
-    vx.v = *reinterpret_cast<const uint4*>(x + i * 8);
-    const unsigned char mb = RELU ? mask[i] : 0xff;
+    vdy.v = *reinterpret_cast<const uint4*>(dy + base);
+    vx.v = *reinterpret_cast<const uint4*>(x + base);
+    const unsigned char mb = RELU ? mask[r * vecC + vec] : 0xff;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      const int c = vc * 8 + k;
       float g = b2f(vdy.h[k]);
       if (RELU && !((mb >> k) & 1)) g = 0.f;
-      float xhat = (b2f(vx.h[k]) - mean[c]) * rstd[c];
-      float d = gamma[c] * rstd[c] * (g - dbeta[c] * invM - xhat * dgamma[c] * invM);
+      float d = A[k] * g - B[k] * b2f(vx.h[k]) + D[k];
       odx.h[k] = f2b(d);
       if (HAS_RES) odr.h[k] = f2b(g);
     }
-    *reinterpret_cast<uint4*>(dx + i * 8) = odx.v;
-    if (HAS_RES) *reinterpret_cast<uint4*>(dres + i * 8) = odr.v;
+    *reinterpret_cast<uint4*>(dx + base) = odx.v;
+    if (HAS_RES) *reinterpret_cast<uint4*>(dres + base) = odr.v;
   }
 }
 
@@ -633,12 +663,27 @@ DDLW_EXPORT int ddlw_bn_finalize(const void* part_sum, const void* part_sumsq,
   DDLW_CHECK_LAUNCH();
 }
 
+
+// grid for the fixed-channel elementwise BN kernels: x covers channel-vector
+// groups, y covers row groups (capped; kernels stride the remainder)
+static inline dim3 bn_ew_grid(long rows, int C) {
+  int vecC = C >> 3;
+  int VPB = vecC < 256 ? vecC : 256;
+  int ROWG = 256 / VPB;
+  int gx = (vecC + VPB - 1) / VPB;
+  long need = (rows + ROWG - 1) / ROWG;
+  long cap = 4096 / gx;
+  if (cap < 1) cap = 1;
+  long gy = need < cap ? need : cap;
+  if (gy < 1) gy = 1;
+  return dim3(gx, (unsigned)gy);
+}
+
 DDLW_EXPORT int ddlw_bn_apply(const void* x, const void* res, void* y,
                               void* mask, const void* mean, const void* rstd,
                               const void* gamma, const void* beta, long rows,
                               int C, int relu, void* stream) {
-  long nvec = rows * (C >> 3);
-  dim3 grid(grid_1d(nvec));
+  dim3 grid = bn_ew_grid(rows, C);
   if (relu && res)
     hipLaunchKernelGGL((k_bn_apply<1, true>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)x, (const bf16_t*)res, (bf16_t*)y,
@@ -701,8 +746,7 @@ DDLW_EXPORT int ddlw_bn_bwd_dx(const void* dy, const void* mask, const void* x,
                                const void* gamma, const void* dbeta,
                                const void* dgamma, void* dx, void* dres,
                                long rows, int C, int relu, void* stream) {
-  long nvec = rows * (C >> 3);
-  dim3 grid(grid_1d(nvec));
+  dim3 grid = bn_ew_grid(rows, C);
   if (relu && dres)
     hipLaunchKernelGGL((k_bn_bwd_dx<1, true>), grid, dim3(256), 0, (hipStream_t)stream,
                        (const bf16_t*)dy, (const unsigned char*)mask, (const bf16_t*)x,
