@@ -351,13 +351,15 @@ __global__ __launch_bounds__(256) void k_maxpool3x3s2_fwd(
     unsigned char* __restrict__ argmax,
     int N, int H, int W, int C, int Ho, int Wo) {
   const int vecC = C >> 3;
-  const long total = (long)N * Ho * Wo * vecC;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    int vc = (int)(i % vecC);
-    long t = i / vecC;
-    int wo = (int)(t % Wo); t /= Wo;
-    int ho = (int)(t % Ho); t /= Ho;
+  // 32-bit index math: launcher guards total < 2^31 (the 64-bit div/mod
+  // chain per grid-stride iteration was a measurable cost on this kernel)
+  const unsigned total = (unsigned)((long)N * Ho * Wo * vecC);
+  for (unsigned i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    unsigned vc = i % (unsigned)vecC;
+    unsigned t = i / (unsigned)vecC;
+    int wo = (int)(t % (unsigned)Wo); t /= (unsigned)Wo;
+    int ho = (int)(t % (unsigned)Ho); t /= (unsigned)Ho;
     int n = (int)t;
     float best[8];
     int bidx[8];
@@ -384,9 +386,14 @@ __global__ __launch_bounds__(256) void k_maxpool3x3s2_fwd(
     bf16x8 o;
     #pragma unroll
     for (int k = 0; k < 8; ++k) o.h[k] = f2b(best[k]);
-    *reinterpret_cast<uint4*>(y + i * 8) = o.v;
-    #pragma unroll
-    for (int k = 0; k < 8; ++k) argmax[i * 8 + k] = (unsigned char)bidx[k];
+    *reinterpret_cast<uint4*>(y + (long)i * 8) = o.v;
+    // one 8-byte store instead of 8 byte-stores
+    uint2 am;
+    am.x = (unsigned)bidx[0] | ((unsigned)bidx[1] << 8) |
+           ((unsigned)bidx[2] << 16) | ((unsigned)bidx[3] << 24);
+    am.y = (unsigned)bidx[4] | ((unsigned)bidx[5] << 8) |
+           ((unsigned)bidx[6] << 16) | ((unsigned)bidx[7] << 24);
+    *reinterpret_cast<uint2*>(argmax + (long)i * 8) = am;
   }
 }
 
@@ -397,13 +404,13 @@ __global__ __launch_bounds__(256) void k_maxpool3x3s2_bwd(
     bf16_t* __restrict__ dx,
     int N, int H, int W, int C, int Ho, int Wo) {
   const int vecC = C >> 3;
-  const long total = (long)N * H * W * vecC;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    int vc = (int)(i % vecC);
-    long t = i / vecC;
-    int w = (int)(t % W); t /= W;
-    int h = (int)(t % H); t /= H;
+  const unsigned total = (unsigned)((long)N * H * W * vecC);
+  for (unsigned i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    unsigned vc = i % (unsigned)vecC;
+    unsigned t = i / (unsigned)vecC;
+    int w = (int)(t % (unsigned)W); t /= (unsigned)W;
+    int h = (int)(t % (unsigned)H); t /= (unsigned)H;
     int n = (int)t;
     float acc[8] = {0};
     // output windows covering (h, w): ho*2-1 <= h <= ho*2+1
@@ -420,16 +427,20 @@ __global__ __launch_bounds__(256) void k_maxpool3x3s2_bwd(
         long obase = (((long)n * Ho + ho) * Wo + wo) * C + (long)vc * 8;
         bf16x8 g;
         g.v = *reinterpret_cast<const uint4*>(dy + obase);
-        unsigned char slot = (unsigned char)(kh * 3 + kw);
+        // one 8-byte argmax load instead of 8 byte-loads
+        uint2 am = *reinterpret_cast<const uint2*>(argmax + obase);
+        unsigned slot = (unsigned)(kh * 3 + kw);
         #pragma unroll
-        for (int k = 0; k < 8; ++k)
-          if (argmax[obase + k] == slot) acc[k] += b2f(g.h[k]);
+        for (int k = 0; k < 8; ++k) {
+          unsigned b = ((k < 4 ? am.x : am.y) >> ((k & 3) * 8)) & 0xffu;
+          if (b == slot) acc[k] += b2f(g.h[k]);
+        }
       }
     }
     bf16x8 o;
     #pragma unroll
     for (int k = 0; k < 8; ++k) o.h[k] = f2b(acc[k]);
-    *reinterpret_cast<uint4*>(dx + i * 8) = o.v;
+    *reinterpret_cast<uint4*>(dx + (long)i * 8) = o.v;
   }
 }
 
@@ -778,6 +789,10 @@ DDLW_EXPORT int ddlw_maxpool3x3s2_fwd(const void* x, void* y, void* argmax,
                                       int N, int H, int W, int C, int Ho, int Wo,
                                       void* stream) {
   long total = (long)N * Ho * Wo * (C >> 3);
+  if ((long)N * H * W * (C >> 3) >= (1ll << 31)) {
+    ddlw_set_error("maxpool3x3s2: tensor too large for 32-bit indexing");
+    return 2;
+  }
   hipLaunchKernelGGL(k_maxpool3x3s2_fwd, dim3(grid_1d(total)), dim3(256), 0,
                      (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)y,
                      (unsigned char*)argmax, N, H, W, C, Ho, Wo);
@@ -788,6 +803,10 @@ DDLW_EXPORT int ddlw_maxpool3x3s2_bwd(const void* dy, const void* argmax, void* 
                                       int N, int H, int W, int C, int Ho, int Wo,
                                       void* stream) {
   long total = (long)N * H * W * (C >> 3);
+  if (total >= (1ll << 31)) {
+    ddlw_set_error("maxpool3x3s2: tensor too large for 32-bit indexing");
+    return 2;
+  }
   hipLaunchKernelGGL(k_maxpool3x3s2_bwd, dim3(grid_1d(total)), dim3(256), 0,
                      (hipStream_t)stream, (const bf16_t*)dy,
                      (const unsigned char*)argmax, (bf16_t*)dx, N, H, W, C, Ho, Wo);
