@@ -37,6 +37,24 @@ def fused_block_enabled() -> bool:
     return os.environ.get("DDLW_FUSED_BLOCK", "1") == "1"
 
 
+def wgrad_stream_enabled() -> bool:
+    """Run the weight-gradient chain of the fused block on a side HIP
+    stream: wgrad GEMMs are independent of the dgrad/BN chain, so they
+    co-run with the small latency-bound finalize kernels that otherwise
+    leave the chip underfilled between the big bandwidth passes."""
+    return os.environ.get("DDLW_WGRAD_STREAM", "1") == "1"
+
+
+_wgrad_streams = {}
+
+
+def _wgrad_stream(device) -> "torch.cuda.Stream":
+    s = _wgrad_streams.get(device)
+    if s is None:
+        s = _wgrad_streams[device] = torch.cuda.Stream(device=device)
+    return s
+
+
 def _cl(t: torch.Tensor) -> torch.Tensor:
     return t.contiguous(memory_format=torch.channels_last)
 
@@ -116,16 +134,48 @@ class _BottleneckFn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
 
+        # wgrads are independent of the dgrad/BN chain: launch them on a
+        # side stream so their GEMMs co-run with the chain's small
+        # latency-bound kernels (grid-underfilled finalizes)
+        use_side = wgrad_stream_enabled() and dy.is_cuda
+        main = torch.cuda.current_stream() if use_side else None
+        side = _wgrad_stream(dy.device) if use_side else None
+        side_evs = []
+
+        def wgrad(dy_t, x_t, w, st, pd, hip_g):
+            if not use_side:
+                _, dw = conv_gemm.conv_backward(dy_t, x_t, w, st, pd,
+                                                False, hip_g, need_dx=False)
+                return dw
+            ready = torch.cuda.Event()
+            ready.record(main)  # dy_t complete on the main stream
+            side.wait_event(ready)
+            with torch.cuda.stream(side):
+                # inputs were allocated on the main stream; tell the caching
+                # allocator they are in use on the side stream
+                dy_t.record_stream(side)
+                x_t.record_stream(side)
+                _, dw = conv_gemm.conv_backward(dy_t, x_t, w, st, pd,
+                                                False, hip_g, need_dx=False)
+                done = torch.cuda.Event()
+                done.record(side)
+            side_evs.append((done, dw))
+            return dw
+
         # bn3 (+residual, +relu): dres is the shortcut-path gradient
         db3, dg3 = binding.bn_bwd_reduce(dy, mask3, t3, m3, s3, True)
         dt3, dres = binding.bn_bwd_dx(dy, mask3, t3, m3, s3, g3, db3, dg3,
                                       True, True)
-        da2, dw3 = conv_gemm.conv_backward(dt3, a2, w3, 1, 0, r3d, r3g)
+        dw3 = wgrad(dt3, a2, w3, 1, 0, r3g)
+        da2, _ = conv_gemm.conv_backward(dt3, a2, w3, 1, 0, r3d, False,
+                                         need_dw=False)
 
         db2, dg2 = binding.bn_bwd_reduce(da2, mask2, t2, m2, s2, True)
         dt2, _ = binding.bn_bwd_dx(da2, mask2, t2, m2, s2, g2, db2, dg2,
                                    True, False)
-        da1, dw2 = conv_gemm.conv_backward(dt2, a1, w2, stride, 1, r2d, r2g)
+        dw2 = wgrad(dt2, a1, w2, stride, 1, r2g)
+        da1, _ = conv_gemm.conv_backward(dt2, a1, w2, stride, 1, r2d, False,
+                                         need_dw=False)
 
         db1, dg1 = binding.bn_bwd_reduce(da1, mask1, t1, m1, s1, True)
         dt1, _ = binding.bn_bwd_dx(da1, mask1, t1, m1, s1, g1, db1, dg1,
@@ -137,12 +187,21 @@ class _BottleneckFn(torch.autograd.Function):
             dbd, dgd = binding.bn_bwd_reduce(dres, None, td, md, sd, False)
             dtd, _ = binding.bn_bwd_dx(dres, None, td, md, sd, gd, dbd, dgd,
                                        False, False)
-            dxd, dwd = conv_gemm.conv_backward(dtd, x, wd, stride, 0, rdd, rdg)
+            dwd = wgrad(dtd, x, wd, stride, 0, rdg)
+            dxd, _ = conv_gemm.conv_backward(dtd, x, wd, stride, 0, rdd, False,
+                                             need_dw=False)
             join = dxd
         else:
             join = dres
+        dw1 = wgrad(dt1, x, w1, 1, 0, r1g)
         # the join-add rides conv1's dgrad epilogue (acc=join)
-        dx, dw1 = conv_gemm.conv_backward(dt1, x, w1, 1, 0, r1d, r1g, acc=join)
+        dx, _ = conv_gemm.conv_backward(dt1, x, w1, 1, 0, r1d, False,
+                                        need_dw=False, acc=join)
+
+        # rejoin: the optimizer (main stream) reads the dw tensors
+        for done, dw in side_evs:
+            main.wait_event(done)
+            dw.record_stream(main)
 
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
                 dwd, dgd, dbd, None)
