@@ -47,8 +47,20 @@ __global__ __launch_bounds__(256) void k_bn_stats(
   float s[8] = {0}, q[8] = {0};
   const long row0 = (long)blockIdx.y * ROWG + rowg;
   const long rstride = (long)gridDim.y * ROWG;
-  if (active)
-    for (long r = row0; r < rows; r += rstride) {
+  if (active) {
+    long r = row0;
+    for (; r + rstride < rows; r += 2 * rstride) {
+      bf16x8 v0, v1;  // two independent strided loads in flight (MLP)
+      v0.v = *reinterpret_cast<const uint4*>(x + r * C + (long)vec * 8);
+      v1.v = *reinterpret_cast<const uint4*>(x + (r + rstride) * C + (long)vec * 8);
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float f0 = b2f(v0.h[k]), f1 = b2f(v1.h[k]);
+        s[k] += f0 + f1;
+        q[k] += f0 * f0 + f1 * f1;
+      }
+    }
+    if (r < rows) {
       bf16x8 v;
       v.v = *reinterpret_cast<const uint4*>(x + r * C + (long)vec * 8);
       #pragma unroll
@@ -58,6 +70,7 @@ __global__ __launch_bounds__(256) void k_bn_stats(
         q[k] += f * f;
       }
     }
+  }
   // reduce across row groups through LDS
   __shared__ float lds[256 * 8];
   if (ROWG > 1) {
@@ -164,7 +177,7 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     shift[k] = beta[c] - mean[c] * scale[k];
   }
   const long rstride = (long)gridDim.y * ROWG;
-  for (long r = (long)blockIdx.y * ROWG + rowg; r < rows; r += rstride) {
+  auto body = [&](long r) {
     const long base = r * C + (long)vec * 8;
     bf16x8 v, o, rv;
     v.v = *reinterpret_cast<const uint4*>(x + base);
@@ -183,7 +196,15 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     *reinterpret_cast<uint4*>(y + base) = o.v;
     if (RELU) mask[r * vecC + vec] = mb;  // 1 byte per 8-channel vector:
                                           // bwd reads this instead of y
+  };
+  // 2x row unroll: two independent strided streams per thread keeps more
+  // loads in flight (single 16-B load/thread was MLP-starved)
+  long r = (long)blockIdx.y * ROWG + rowg;
+  for (; r + rstride < rows; r += 2 * rstride) {
+    body(r);
+    body(r + rstride);
   }
+  if (r < rows) body(r);
 }
 
 // ---------------------------------------------------------------------------
@@ -214,7 +235,7 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
     for (int k = 0; k < 8; ++k) { mn[k] = mean[vec * 8 + k]; rs[k] = rstd[vec * 8 + k]; }
     const long row0 = (long)blockIdx.y * ROWG + rowg;
     const long rstride = (long)gridDim.y * ROWG;
-    for (long r = row0; r < rows; r += rstride) {
+    auto body = [&](long r) {
       const long base = r * C + (long)vec * 8;
       bf16x8 vdy, vx;
       vdy.v = *reinterpret_cast<const uint4*>(dy + base);
@@ -227,7 +248,13 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
         db[k] += g;
         dg[k] += g * (b2f(vx.h[k]) - mn[k]) * rs[k];
       }
+    };
+    long r = row0;
+    for (; r + rstride < rows; r += 2 * rstride) {
+      body(r);
+      body(r + rstride);
     }
+    if (r < rows) body(r);
   }
   __shared__ float lds[256 * 8];
   if (ROWG > 1) {
@@ -322,7 +349,7 @@ __global__ __launch_bounds__(256) void k_bn_bwd_dx(
     D[k] = B[k] * mean[c] - A[k] * dbeta[c] * invM;
   }
   const long rstride = (long)gridDim.y * ROWG;
-  for (long r = (long)blockIdx.y * ROWG + rowg; r < rows; r += rstride) {
+  auto body = [&](long r) {
     const long base = r * C + (long)vec * 8;
     bf16x8 vdy, vx, odx, odr;
     vdy.v = *reinterpret_cast<const uint4*>(dy + base);
@@ -338,7 +365,13 @@ __global__ __launch_bounds__(256) void k_bn_bwd_dx(
     }
     *reinterpret_cast<uint4*>(dx + base) = odx.v;
     if (HAS_RES) *reinterpret_cast<uint4*>(dres + base) = odr.v;
+  };
+  long r = (long)blockIdx.y * ROWG + rowg;
+  for (; r + rstride < rows; r += 2 * rstride) {
+    body(r);
+    body(r + rstride);
   }
+  if (r < rows) body(r);
 }
 
 // ---------------------------------------------------------------------------
