@@ -51,3 +51,25 @@ def test_model_save_load_roundtrip(ddlw_home):
     m3 = load_model("models:/cnn/production")
     m3.eval()
     assert torch.allclose(m(x), m3(x), atol=1e-6)
+
+
+def test_fused_block_gating_cpu(monkeypatch):
+    """bottleneck_fusable must refuse CPU/eval/disabled configurations
+    (the fused Function is a GPU-training-only path)."""
+    import torch
+
+    from ddlw_amd.models.resnet import Bottleneck
+    from ddlw_amd.ops import block as B
+
+    blk = Bottleneck(256, 64)
+    x = torch.randn(2, 256, 14, 14)
+    blk.train()
+    assert not B.bottleneck_fusable(blk, x)  # CPU tensor
+    monkeypatch.setenv("DDLW_FUSED_BLOCK", "0")
+    assert not B.bottleneck_fusable(blk, x)  # disabled
+    monkeypatch.delenv("DDLW_FUSED_BLOCK")
+    blk.eval()
+    assert not B.bottleneck_fusable(blk, x)  # eval mode
+    blk.train()
+    with torch.no_grad():
+        assert not B.bottleneck_fusable(blk, x)  # grad disabled
