@@ -235,7 +235,9 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
     for (int k = 0; k < 8; ++k) { mn[k] = mean[vec * 8 + k]; rs[k] = rstd[vec * 8 + k]; }
     const long row0 = (long)blockIdx.y * ROWG + rowg;
     const long rstride = (long)gridDim.y * ROWG;
-    auto body = [&](long r) {
+    // (2x unroll measured slower here — two dy+x+mask streams thrash;
+    // the read-only stats kernel keeps its unroll, this one stays simple)
+    for (long r = row0; r < rows; r += rstride) {
       const long base = r * C + (long)vec * 8;
       bf16x8 vdy, vx;
       vdy.v = *reinterpret_cast<const uint4*>(dy + base);
@@ -248,13 +250,7 @@ __global__ __launch_bounds__(256) void k_bn_bwd_reduce(
         db[k] += g;
         dg[k] += g * (b2f(vx.h[k]) - mn[k]) * rs[k];
       }
-    };
-    long r = row0;
-    for (; r + rstride < rows; r += 2 * rstride) {
-      body(r);
-      body(r + rstride);
     }
-    if (r < rows) body(r);
   }
   __shared__ float lds[256 * 8];
   if (ROWG > 1) {
