@@ -69,6 +69,13 @@ class FusedSGD(torch.optim.Optimizer):
         self._desc[gi] = (key, desc, max_numel, cpu)
         return desc, max_numel
 
+    def load_state_dict(self, state_dict):
+        # state tensors (master/momentum) are replaced by new storage, but
+        # the cached chunk descriptors hold their raw device pointers and
+        # the cache key only tracks param/grad pointers -> must invalidate
+        super().load_state_dict(state_dict)
+        self._desc.clear()
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = closure() if closure is not None else None
@@ -151,6 +158,13 @@ class FusedAdam(torch.optim.Optimizer):
         # async H2D may still be reading it when this function returns)
         self._desc[gi] = (key, desc, max_numel, cpu)
         return desc, max_numel
+
+    def load_state_dict(self, state_dict):
+        # state tensors (master/m/v) are replaced by new storage, but
+        # the cached chunk descriptors hold their raw device pointers and
+        # the cache key only tracks param/grad pointers -> must invalidate
+        super().load_state_dict(state_dict)
+        self._desc.clear()
 
     @torch.no_grad()
     def step(self, closure=None):

@@ -349,3 +349,4 @@ def test_training_equivalence_hip_vs_stock():
     assert hip[-1] < hip[0] and ref[-1] < ref[0], (hip, ref)
     for a, b in zip(hip, ref):
         assert abs(a - b) / max(abs(b), 0.3) < 0.35, (hip, ref)
+

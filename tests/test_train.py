@@ -131,3 +131,20 @@ def test_checkpoint_resume(ddlw_home, tmp_path):
         assert torch.allclose(v1, v2), k1
     hist = m2.fit(_toy_data(), epochs=1, verbose=0)
     assert len(hist.history["loss"]) == 1
+
+
+def test_fused_optimizer_desc_invalidated_on_state_load():
+    """load_state_dict replaces state tensors; the chunk-descriptor cache
+    must be invalidated or the kernel would write through stale pointers."""
+    import torch
+
+    from ddlw_amd.ops.optim import FusedAdam, FusedSGD
+
+    for cls in (FusedSGD, FusedAdam):
+        p = torch.nn.Parameter(torch.randn(32))
+        opt = cls([p])
+        p.grad = torch.randn(32)
+        opt.step()
+        opt._desc[0] = (("sentinel",), None, 0)  # simulate a built cache
+        opt.load_state_dict(opt.state_dict())
+        assert opt._desc == {}, cls.__name__
