@@ -173,3 +173,20 @@ def test_runner_no_restart_still_aborts(tmp_path):
         Runner(np=2, timeout_s=60, max_restarts=0).run(
             _elastic_worker, ckpt_dir=str(tmp_path)
         )
+
+
+def _hang_worker():
+    import time as _t
+
+    if api.rank() == 1:
+        _t.sleep(60)
+    return True
+
+
+def test_runner_timeout_aborts_hung_gang():
+    import time as _t
+
+    t0 = _t.time()
+    with pytest.raises((TimeoutError, RuntimeError)):
+        Runner(np=2, timeout_s=6).run(_hang_worker)
+    assert _t.time() - t0 < 45  # aborted well before the 60s hang
