@@ -50,9 +50,13 @@ def init(backend: Optional[str] = None, timeout_s: float = 300.0) -> None:
     _world_size = world
     if world > 1:
         if backend is None:
-            backend = os.environ.get("DDLW_BACKEND") or (
-                "nccl" if torch.cuda.is_available() else "gloo"
-            )
+            backend = os.environ.get("DDLW_BACKEND")
+            if backend is None:
+                ngpu = torch.cuda.device_count() if torch.cuda.is_available() else 0
+                # RCCL refuses two ranks on one physical GPU — oversubscribed
+                # worlds (e.g. np=2 smoke runs on a 1-GPU box) ride gloo,
+                # which moves CUDA tensors via host staging
+                backend = "nccl" if ngpu >= world else "gloo"
         if torch.cuda.is_available():
             torch.cuda.set_device(_local_rank % torch.cuda.device_count())
         import datetime
