@@ -263,10 +263,19 @@ def fmin(
             )
             if verbose:
                 loss = result.get("loss")
-                print(
-                    f"[fmin] trial {len(trials.trials)}/{max_evals} "
-                    f"loss={loss if loss is not None else 'FAIL'}",
-                    flush=True,
-                )
+                if loss is None:
+                    err = (result.get("error") or "").strip().splitlines()
+                    detail = f" ({err[-1]})" if err else ""
+                    print(
+                        f"[fmin] trial {len(trials.trials)}/{max_evals} "
+                        f"loss=FAIL{detail}",
+                        flush=True,
+                    )
+                else:
+                    print(
+                        f"[fmin] trial {len(trials.trials)}/{max_evals} "
+                        f"loss={loss}",
+                        flush=True,
+                    )
     best = trials.best_trial
     return dict(best["params"])
