@@ -55,6 +55,9 @@ def setup(root: Optional[str] = None, user: Optional[str] = None) -> SetupCfg:
     kw = {}
     if root is not None:
         kw["root"] = Path(root)
+        # export so worker subprocesses (trials, ranks, UDF workers) resolve
+        # the same data root (mirrors DDLW_TRACKING_URI propagation)
+        os.environ["DDLW_HOME"] = str(root)
     if user is not None:
         kw["user"] = user
     if _SETUP is None or kw:

@@ -56,3 +56,17 @@ def test_preprocess_range_and_parity(jpeg_tree, ddlw_home):
     # str-typed content workaround path (reference P2/03:228-229)
     p2 = preprocess_pil(str(jpg), 24, 24)
     assert (p2 == p).all()
+
+
+def test_setup_root_propagates_to_env(tmp_path, monkeypatch):
+    """setup(root=...) must export DDLW_HOME so trial/rank subprocesses
+    resolve the same warehouse (regression: GPU-box HPO trials failed with
+    FileNotFoundError on the default root)."""
+    import os
+
+    import ddlw_amd.core.config as config
+
+    config._SETUP = None
+    config.setup(root=str(tmp_path / "r1"))
+    assert os.environ["DDLW_HOME"] == str(tmp_path / "r1")
+    config._SETUP = None
