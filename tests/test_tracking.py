@@ -68,3 +68,30 @@ def test_registry_stage_transitions(ddlw_home):
 def test_filter_parse_errors(ddlw_home):
     with pytest.raises(ValueError):
         tracking._parse_filter("malformed ~~ clause")
+
+
+def test_search_runs_missing_order_key(ddlw_home):
+    """Reference quirk 3: ordering by a metric some runs
+    never logged must not crash — absent keys sort last, present ones win."""
+    from ddlw_amd.core import tracking
+
+    tracking.set_experiment("order_edge")
+    with tracking.start_run(run_name="a") as r:
+        r.log_metric("val_loss", 0.5)        # no 'accuracy' at all
+    with tracking.start_run(run_name="b") as r:
+        r.log_metric("accuracy", 0.9)
+    df = tracking.search_runs(order_by=["metrics.accuracy DESC"])
+    assert len(df) == 2
+    assert df.iloc[0]["metrics.accuracy"] == 0.9  # the run that HAS the key
+
+
+def test_search_runs_like_operator(ddlw_home):
+    from ddlw_amd.core import tracking
+
+    tracking.set_experiment("like_edge")
+    with tracking.start_run(run_name="trial-1") as r:
+        r.set_tag("kind", "hpo")
+    with tracking.start_run(run_name="other") as r:
+        r.set_tag("kind", "manual")
+    df = tracking.search_runs(filter_string="tags.kind = 'hpo'")
+    assert len(df) == 1
