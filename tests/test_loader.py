@@ -97,3 +97,25 @@ def test_loader_label_parity(ddlw_home):
     expect = tbl.column("label_idx").to_pylist()
     assert labels.tolist() == expect
     conv.delete()
+
+
+def test_loader_row_level_fallback_more_ranks_than_groups(tmp_path):
+    """Fewer row groups than ranks: row-level round-robin keeps every shard
+    non-empty, disjoint, exhaustive (the hang-prevention path)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from ddlw_amd.data.loader import ShardedParquetLoader
+
+    t = pa.table({"content": [b"x"] * 10, "label_idx": list(range(10))})
+    pq.write_table(t, tmp_path / "p.parquet", row_group_size=10)  # ONE group
+    seen = []
+    for r in range(4):
+        ld = ShardedParquetLoader(
+            str(tmp_path), batch_size=3, cur_shard=r, shard_count=4,
+            num_epochs=1, transform=lambda c: __import__("torch").zeros(1),
+        )
+        labels = [l for _, l in ld._iter_rows()]
+        assert labels, f"rank {r} got an empty shard"
+        seen.extend(labels)
+    assert sorted(seen) == list(range(10))  # disjoint + exhaustive
