@@ -181,7 +181,7 @@ def broadcast_optimizer_state(optimizer: torch.optim.Optimizer, root_rank: int =
 
 
 class _Bucket:
-    __slots__ = ("params", "bytes", "flat", "work", "ev", "ready")
+    __slots__ = ("params", "bytes", "flat", "work", "ev", "ready", "live")
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
@@ -190,6 +190,7 @@ class _Bucket:
         self.work = None
         self.ev = None  # native-path completion event
         self.ready = 0
+        self.live: List[torch.nn.Parameter] = []  # params with grads this step
 
 
 class DistributedOptimizer:
@@ -254,7 +255,12 @@ class DistributedOptimizer:
         return hook
 
     def _launch(self, bucket: _Bucket) -> None:
-        grads = [p.grad for p in bucket.params]
+        # params consistently unused across ranks (no grad) stay out of the
+        # flat buffer; ranks must agree on graph structure (DDP contract)
+        bucket.live = [p for p in bucket.params if p.grad is not None]
+        if not bucket.live:
+            return
+        grads = [p.grad for p in bucket.live]
         flat = torch._utils._flatten_dense_tensors(grads)
         if self._tracer.enabled:
             import time as _t
@@ -297,8 +303,9 @@ class DistributedOptimizer:
         if not _pg_active():
             return
         for b in self._buckets:
-            if b.flat is None and b.params and b.params[0].grad is not None:
-                # hook missed (e.g. grads produced outside autograd): reduce now
+            if b.flat is None and any(p.grad is not None for p in b.params):
+                # hook missed (grads produced outside autograd, or a bucket
+                # containing unused params never hit its ready count)
                 self._launch(b)
         for b in self._buckets:
             if b.flat is None:
@@ -312,9 +319,9 @@ class DistributedOptimizer:
                 b.ev = None
             elif self.average:
                 b.flat.div_(_world_size)
-            grads = [p.grad for p in b.params]
+            grads = [p.grad for p in b.live]
             for p, g in zip(
-                b.params, torch._utils._unflatten_dense_tensors(b.flat, grads)
+                b.live, torch._utils._unflatten_dense_tensors(b.flat, grads)
             ):
                 p.grad.copy_(g)
             b.flat = None

@@ -190,3 +190,30 @@ def test_runner_timeout_aborts_hung_gang():
     with pytest.raises((TimeoutError, RuntimeError)):
         Runner(np=2, timeout_s=6).run(_hang_worker)
     assert _t.time() - t0 < 45  # aborted well before the 60s hang
+
+
+def _unused_param_worker(seed):
+    """Model with a parameter that never receives a gradient (unused head):
+    the bucketed optimizer must reduce the live grads and leave the unused
+    param's grad None — without crashing or silently skipping buckets."""
+    torch.manual_seed(seed)
+    lin = torch.nn.Linear(8, 4)
+    unused = torch.nn.Parameter(torch.randn(3))
+    params = [unused, lin.weight, lin.bias]
+    opt = api.DistributedOptimizer(
+        torch.optim.SGD(params, lr=0.1), bucket_cap_mb=0.001
+    )
+    api.broadcast_parameters(params)
+    x = torch.randn(4, 8) + api.rank()
+    lin(x).sum().backward()
+    opt.step()
+    return {
+        "unused_grad_none": unused.grad is None,
+        "w": lin.weight.detach().clone(),
+    }
+
+
+def test_distributed_optimizer_with_unused_param():
+    out = Runner(np=2, timeout_s=120).run(_unused_param_worker, seed=3)
+    assert out["unused_grad_none"]
+    assert torch.isfinite(out["w"]).all()
