@@ -65,8 +65,11 @@ def _make_optimizer(spec, params, lr: Optional[float] = None):
 
 
 def sparse_categorical_crossentropy_from_logits(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-    """The reference's loss (``.../02_model_training_single_node.py:202``)."""
-    return F.cross_entropy(logits.float(), labels)
+    """The reference's loss (``.../02_model_training_single_node.py:202``);
+    fused HIP kernel (K9) on GPU, stock cross_entropy on CPU."""
+    from ..ops.layers import softmax_cross_entropy
+
+    return softmax_cross_entropy(logits, labels)
 
 
 class History:
