@@ -143,12 +143,15 @@ def test_gap_parity():
     assert torch.allclose(xb.grad.float(), x32.grad, atol=2e-2, rtol=2e-2)
 
 
-def test_softmax_ce_parity():
+@pytest.mark.parametrize("C", [5, 33, 1000])
+def test_softmax_ce_parity(C):
+    """C=5 is the reference's class count; 33 exercises the partial-wave
+    column tail; 1000 the bench config."""
     from ddlw_amd.ops.layers import softmax_cross_entropy
 
     torch.manual_seed(11)
-    logits = torch.randn(64, 1000, device=_cuda()) * 4
-    labels = torch.randint(0, 1000, (64,), device=_cuda())
+    logits = torch.randn(64, C, device=_cuda()) * 4
+    labels = torch.randint(0, C, (64,), device=_cuda())
     l32 = logits.detach().requires_grad_(True)
     lb = logits.detach().requires_grad_(True)
     loss = softmax_cross_entropy(lb, labels)
