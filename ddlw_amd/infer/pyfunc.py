@@ -208,7 +208,18 @@ class PredictUDF:
             self._procs.append(p)
         ready = 0
         while ready < num_workers:
-            status, _, payload = self._result_q.get()
+            import queue as _q
+
+            try:
+                status, _, payload = self._result_q.get(timeout=5.0)
+            except _q.Empty:
+                dead = [p.pid for p in self._procs if not p.is_alive()]
+                if dead:
+                    self.close()
+                    raise RuntimeError(
+                        f"predict worker(s) died during model load (pids {dead})"
+                    )
+                continue
             if status == "err":
                 self.close()
                 raise RuntimeError(f"predict worker failed to load model:\n{payload}")
@@ -225,8 +236,22 @@ class PredictUDF:
             self._task_qs[i].put((i, part))
         results: List[Optional[List[str]]] = [None] * n
         errs = []
-        for _ in range(n):
-            status, task_id, payload = self._result_q.get()
+        got = 0
+        while got < n:
+            # liveness-checked wait: a worker dying mid-task (OOM, kill)
+            # must surface as an error, not a hang
+            import queue as _q
+
+            try:
+                status, task_id, payload = self._result_q.get(timeout=5.0)
+            except _q.Empty:
+                dead = [p.pid for p in self._procs if not p.is_alive()]
+                if dead:
+                    raise RuntimeError(
+                        f"predict worker(s) died mid-task (pids {dead})"
+                    )
+                continue
+            got += 1
             if status == "ok":
                 results[task_id] = payload
             else:

@@ -77,3 +77,32 @@ def test_predict_udf_matches_single_node(ddlw_home):
     udf = predict_udf(uri, num_workers=3, gpus=[])
     fanned = udf(contents)
     assert fanned == [str(s) for s in single]
+
+
+class _DyingModel(PythonModel):
+    """Predict kills the worker process — the pool must raise, not hang."""
+
+    def load_context(self, context):
+        pass
+
+    def predict(self, context, rows):
+        import os as _os
+
+        _os._exit(41)
+
+
+def test_predict_udf_detects_dead_worker(ddlw_home):
+    import time
+
+    import pytest
+
+    from ddlw_amd.infer.pyfunc import log_model, predict_udf
+
+    tracking.set_experiment("udf_death")
+    with tracking.start_run():
+        uri = log_model("killer", _DyingModel())
+    t0 = time.time()
+    with predict_udf(uri, num_workers=1) as udf:
+        with pytest.raises(RuntimeError, match="died"):
+            udf([b"row"])
+    assert time.time() - t0 < 60
