@@ -81,9 +81,11 @@ class LocalTrials(Trials):
     as subprocesses, each pinned to one GPU (HIP_VISIBLE_DEVICES rotation).
     """
 
-    def __init__(self, parallelism: int = 4, gpus: Optional[List[int]] = None):
+    def __init__(self, parallelism: int = 4, gpus: Optional[List[int]] = None,
+                 trial_timeout_s: Optional[float] = None):
         super().__init__()
         self.parallelism = max(1, parallelism)
+        self.trial_timeout_s = trial_timeout_s  # None = unbounded (training)
         if gpus is None:
             try:
                 import torch
@@ -117,7 +119,17 @@ class LocalTrials(Trials):
             p.start()
             procs.append((i, p, q))
         for i, p, q in procs:
-            p.join()
+            p.join(self.trial_timeout_s)
+            if p.is_alive():
+                # hung trial: kill it and record a failure instead of
+                # blocking the whole sweep (SURVEY.md §5.3 discipline)
+                p.terminate()
+                p.join(5)
+                results[i] = {
+                    "status": STATUS_FAIL,
+                    "error": f"trial timed out after {self.trial_timeout_s}s",
+                }
+                continue
             results[i] = (
                 q.get()
                 if not q.empty()

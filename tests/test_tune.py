@@ -86,3 +86,23 @@ def test_failed_trials_recorded(ddlw_home):
     with pytest.raises(RuntimeError):
         fmin(_failing_obj, {"x": hp.uniform("x", 0, 1)}, max_evals=3, trials=trials, verbose=False)
     assert all(t["result"]["status"] == "fail" for t in trials.trials)
+
+
+def _sleepy_objective(params):
+    import time
+
+    time.sleep(120)
+    return 0.0
+
+
+def test_local_trials_timeout(ddlw_home):
+    """A hung trial is killed and recorded as FAIL, not awaited forever."""
+    import time
+
+    from ddlw_amd.tune.fmin import STATUS_FAIL, LocalTrials
+
+    t0 = time.time()
+    trials = LocalTrials(parallelism=1, gpus=[], trial_timeout_s=5)
+    res = trials.run_batch(_sleepy_objective, [{"x": 1.0}])
+    assert res[0]["status"] == STATUS_FAIL and "timed out" in res[0]["error"]
+    assert time.time() - t0 < 60
