@@ -36,7 +36,10 @@ class Conv2d(nn.Conv2d):
             and self.groups == self.in_channels == self.out_channels
             and self.in_channels % 8 == 0
             and self.bias is None
-            and not (x.requires_grad or self.weight.requires_grad)
+            and (
+                not torch.is_grad_enabled()  # inference: fwd-only kernel safe
+                or not (x.requires_grad or self.weight.requires_grad)
+            )
         ):
             # depthwise (K2 — MobileNetV2 blocks), inference/frozen path
             from . import binding
