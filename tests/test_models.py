@@ -73,3 +73,24 @@ def test_fused_block_gating_cpu(monkeypatch):
     blk.train()
     with torch.no_grad():
         assert not B.bottleneck_fusable(blk, x)  # grad disabled
+
+
+def test_conv_backward_helper_cpu_fallback_with_acc():
+    """The shared conv_backward helper's library fallback must apply the
+    accumulate input (the fused block's join-add) identically to autograd."""
+    import torch
+    import torch.nn.functional as F
+
+    from ddlw_amd.ops.conv_gemm import conv_backward
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 4, 8, 8, requires_grad=True)
+    w = torch.randn(6, 4, 3, 3, requires_grad=True)
+    acc = torch.randn(2, 4, 8, 8)
+    y = F.conv2d(x, w, None, 1, 1)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    dx, dw = conv_backward(dy, x.detach(), w.detach(), 1, 1,
+                           hip_dgrad=False, hip_wgrad=False, acc=acc)
+    assert torch.allclose(dx.float(), (x.grad + acc).float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(dw.float(), w.grad.float(), atol=2e-2, rtol=2e-2)
