@@ -16,11 +16,48 @@ is one kernel launch per group.
 """
 from __future__ import annotations
 
+from collections import defaultdict
+from copy import deepcopy
 from typing import Dict, List, Tuple
 
 import torch
 
 from . import binding
+
+
+def _load_state_dict_no_cast(opt: torch.optim.Optimizer, state_dict) -> None:
+    """Load optimizer state WITHOUT torch's param-dtype casting.
+
+    torch.optim.Optimizer.load_state_dict casts floating-point state tensors
+    to the param's dtype — for bf16 params that turns the fp32 master /
+    momentum / exp_avg buffers into bf16, but the fused kernels read those
+    buffers through raw fp32 pointers (out-of-bounds reads + precision loss).
+    This loader maps saved state to params positionally and only moves
+    tensors to the param's device, preserving dtype exactly.
+    """
+    sd = deepcopy(state_dict)
+    groups = opt.param_groups
+    saved_groups = sd["param_groups"]
+    if len(groups) != len(saved_groups):
+        raise ValueError("loaded state dict has a different number of parameter groups")
+    id_map: Dict[int, torch.Tensor] = {}
+    for g, sg in zip(groups, saved_groups):
+        if len(g["params"]) != len(sg["params"]):
+            raise ValueError("loaded state dict contains a parameter group that "
+                             "doesn't match the size of optimizer's group")
+        id_map.update(zip(sg["params"], g["params"]))
+    new_state: Dict[torch.Tensor, dict] = {}
+    for k, v in sd["state"].items():
+        p = id_map[k]
+        new_state[p] = {
+            kk: (vv.to(p.device) if isinstance(vv, torch.Tensor) else vv)
+            for kk, vv in v.items()
+        }
+    opt.state = defaultdict(dict, new_state)
+    for g, sg in zip(groups, saved_groups):
+        for kk, vv in sg.items():
+            if kk != "params":
+                g[kk] = vv
 
 
 class FusedSGD(torch.optim.Optimizer):
@@ -70,10 +107,10 @@ class FusedSGD(torch.optim.Optimizer):
         return desc, max_numel
 
     def load_state_dict(self, state_dict):
-        # state tensors (master/momentum) are replaced by new storage, but
-        # the cached chunk descriptors hold their raw device pointers and
-        # the cache key only tracks param/grad pointers -> must invalidate
-        super().load_state_dict(state_dict)
+        # no-cast load (fp32 master/momentum of bf16 params must stay fp32
+        # for the raw-pointer fused kernel); also invalidate the descriptor
+        # cache — it holds raw device pointers into the replaced state
+        _load_state_dict_no_cast(self, state_dict)
         self._desc.clear()
 
     @torch.no_grad()
@@ -159,11 +196,25 @@ class FusedAdam(torch.optim.Optimizer):
         self._desc[gi] = (key, desc, max_numel, cpu)
         return desc, max_numel
 
+    def state_dict(self):
+        # persist the shared step count per-param under 'step' so a fresh
+        # instance resumes bias correction at the right t (copy the inner
+        # dicts — super() returns references into live state)
+        sd = super().state_dict()
+        sd["state"] = {k: dict(v) for k, v in sd["state"].items()}
+        for st in sd["state"].values():
+            st["step"] = self._step_t
+        return sd
+
     def load_state_dict(self, state_dict):
-        # state tensors (master/m/v) are replaced by new storage, but
-        # the cached chunk descriptors hold their raw device pointers and
-        # the cache key only tracks param/grad pointers -> must invalidate
-        super().load_state_dict(state_dict)
+        # no-cast load (fp32 master/m/v of bf16 params must stay fp32 for
+        # the raw-pointer fused kernel); restore the step count; invalidate
+        # the descriptor cache holding raw pointers into the replaced state
+        _load_state_dict_no_cast(self, state_dict)
+        self._step_t = max(
+            (st.get("step", 0) for st in self.state.values()), default=0)
+        for st in self.state.values():
+            st.pop("step", None)
         self._desc.clear()
 
     @torch.no_grad()

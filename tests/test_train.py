@@ -148,3 +148,58 @@ def test_fused_optimizer_desc_invalidated_on_state_load():
         opt._desc[0] = (("sentinel",), None, 0)  # simulate a built cache
         opt.load_state_dict(opt.state_dict())
         assert opt._desc == {}, cls.__name__
+
+
+def test_fused_optimizer_fresh_instance_resume_bf16():
+    """resuming a fused optimizer on a FRESH instance must
+    (a) keep the fp32 master/momentum/m/v state fp32 for bf16 params (torch's
+    default load casts them to the param dtype) and (b) restore Adam's step
+    count so bias correction doesn't restart at t=1. Checked by comparing a
+    save/load-interrupted run against an uninterrupted one."""
+    import torch
+
+    from ddlw_amd.ops.optim import FusedAdam, FusedSGD
+
+    def make(cls, seed):
+        torch.manual_seed(seed)
+        p32 = torch.nn.Parameter(torch.randn(17))
+        p16 = torch.nn.Parameter(torch.randn(23).to(torch.bfloat16))
+        return p32, p16, cls([p32, p16], lr=0.05)
+
+    for cls in (FusedSGD, FusedAdam):
+        torch.manual_seed(7)
+        grads = [(torch.randn(17), torch.randn(23)) for _ in range(6)]
+
+        # uninterrupted run
+        a32, a16, opt_a = make(cls, 0)
+        for g32, g16 in grads:
+            a32.grad = g32.clone()
+            a16.grad = g16.to(torch.bfloat16)
+            opt_a.step()
+
+        # interrupted at step 3: save, rebuild fresh, load, continue
+        b32, b16, opt_b = make(cls, 0)
+        for g32, g16 in grads[:3]:
+            b32.grad = g32.clone()
+            b16.grad = g16.to(torch.bfloat16)
+            opt_b.step()
+        sd = opt_b.state_dict()
+        c32, c16, opt_c = make(cls, 0)
+        with torch.no_grad():
+            c32.copy_(b32)
+            c16.copy_(b16)
+        opt_c.load_state_dict(sd)
+        # state dtype preserved (the fused kernels read these as fp32)
+        for st in opt_c.state.values():
+            for k, v in st.items():
+                if isinstance(v, torch.Tensor):
+                    assert v.dtype == torch.float32, k
+        if cls is FusedAdam:
+            assert opt_c._step_t == 3
+        for g32, g16 in grads[3:]:
+            c32.grad = g32.clone()
+            c16.grad = g16.to(torch.bfloat16)
+            opt_c.step()
+
+        assert torch.allclose(a32, c32, atol=1e-6), cls.__name__
+        assert torch.equal(a16, c16), cls.__name__
