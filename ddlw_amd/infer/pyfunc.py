@@ -229,9 +229,23 @@ class PredictUDF:
         rows = list(rows)
         if not rows:
             return []
+        # drain results stranded by a previous failed call (worker died
+        # mid-task -> we raised with results still in flight); stale task
+        # ids must not be attributed to this call
+        import queue as _q
+
+        while True:
+            try:
+                self._result_q.get_nowait()
+            except _q.Empty:
+                break
         n = min(self.num_workers, len(rows))
         chunk = (len(rows) + n - 1) // n
-        parts = [rows[i * chunk : (i + 1) * chunk] for i in range(n)]
+        # ceil-chunking can produce trailing EMPTY parts on uneven splits
+        # (e.g. 9 rows / 8 workers -> 2,2,2,2,1,0,0,0); drop them so no
+        # worker is asked to predict([])
+        parts = [p for p in (rows[i * chunk : (i + 1) * chunk] for i in range(n)) if p]
+        n = len(parts)
         for i, part in enumerate(parts):
             self._task_qs[i].put((i, part))
         results: List[Optional[List[str]]] = [None] * n

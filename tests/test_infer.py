@@ -106,3 +106,16 @@ def test_predict_udf_detects_dead_worker(ddlw_home):
         with pytest.raises(RuntimeError, match="died"):
             udf([b"row"])
     assert time.time() - t0 < 60
+
+
+def test_predict_udf_uneven_split_no_empty_parts(ddlw_home):
+    """5 rows on 4 workers must not dispatch empty chunks
+    (models like np.stack-based ones raise on empty input)."""
+    uri = _package(ddlw_home)
+    contents, _ = make_synthetic_dataset(5, 16, 16, num_classes=5, seed=3, jpeg=True)
+    single = list(load_model(uri).predict(contents))
+    with predict_udf(uri, num_workers=4, gpus=[]) as udf:
+        fanned = udf(contents)
+        assert fanned == [str(s) for s in single]
+        # reuse the pool: a second call must be unaffected
+        assert udf(contents[:2]) == [str(s) for s in single[:2]]
