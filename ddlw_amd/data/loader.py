@@ -23,13 +23,17 @@ MI355X-native design (not a Petastorm port):
 """
 from __future__ import annotations
 
+import heapq
+import itertools
 import queue
 import threading
 import uuid
+from collections import deque
 from concurrent.futures import ThreadPoolExecutor
 from pathlib import Path
 from typing import Callable, Iterator, List, Optional, Tuple
 
+import numpy as np
 import pyarrow as pa
 import pyarrow.parquet as pq
 import torch
@@ -44,6 +48,106 @@ def shard_row_groups(num_row_groups: int, cur_shard: int, shard_count: int) -> L
     if not (0 <= cur_shard < shard_count):
         raise ValueError(f"cur_shard {cur_shard} out of range for {shard_count}")
     return list(range(cur_shard, num_row_groups, shard_count))
+
+
+def _decode_worker(transform, shm_name, view_shape, dtype_str, task_q, res_q):
+    """Decode-pool worker: pull (seq, slot, [jpeg bytes...]) tasks, decode
+    each image with ``transform`` straight into shared-memory slot ``slot``,
+    report (status, seq, slot, n). Decoded pixels never cross a pipe."""
+    from multiprocessing import shared_memory
+
+    shm = shared_memory.SharedMemory(name=shm_name)
+    view = np.ndarray(view_shape, dtype=np.dtype(dtype_str), buffer=shm.buf)
+    try:
+        while True:
+            task = task_q.get()
+            if task is None:
+                return
+            seq, slot, chunk = task
+            try:
+                for i, c in enumerate(chunk):
+                    out = transform(c)
+                    if isinstance(out, torch.Tensor):
+                        out = out.numpy()
+                    view[slot, i] = out
+                res_q.put(("ok", seq, slot, len(chunk)))
+            except Exception as e:  # surfaced in the main process
+                res_q.put(("err", seq, slot, repr(e)))
+    finally:
+        shm.close()
+
+
+class _ProcDecodePool:
+    """Process-based JPEG decode pool over a shared-memory slot ring.
+
+    The GIL makes a thread pool top out near single-core decode throughput
+    (round-1 measured 1,240 img/s with 96 threads vs a 8,900 img/s GPU step);
+    real parallel decode needs processes. Workers are forked (the transform
+    callable is inherited, no pickling) and write decoded images directly
+    into /dev/shm slots sized one batch each; the main process only ships
+    jpeg bytes in and slot indices out, then memcpys slot -> pinned buffer
+    (GIL released) for the side-stream H2D. Petastorm's reader-pool
+    equivalent (reference P1/03:199-200)."""
+
+    SHM_CAP_BYTES = 8 << 30  # ring cap; also caps in-flight parallelism
+
+    def __init__(self, transform, sample_shape, sample_dtype, batch_size: int,
+                 workers: int):
+        import multiprocessing as mp
+        from multiprocessing import shared_memory
+
+        method = "fork" if "fork" in mp.get_all_start_methods() else "spawn"
+        ctx = mp.get_context(method)
+        self.workers = max(1, int(workers))
+        self.slot_shape = (int(batch_size),) + tuple(int(s) for s in sample_shape)
+        self.np_dtype = np.dtype(sample_dtype)
+        slot_bytes = int(np.prod(self.slot_shape)) * self.np_dtype.itemsize
+        self.slots = max(2, min(self.workers + 2,
+                                self.SHM_CAP_BYTES // max(slot_bytes, 1)))
+        self.shm = shared_memory.SharedMemory(
+            create=True, size=max(slot_bytes * self.slots, 1))
+        view_shape = (self.slots,) + self.slot_shape
+        self.view = np.ndarray(view_shape, dtype=self.np_dtype, buffer=self.shm.buf)
+        self.task_q = ctx.Queue()
+        self.res_q = ctx.Queue()
+        self.procs = [
+            ctx.Process(
+                target=_decode_worker,
+                args=(transform, self.shm.name, view_shape, self.np_dtype.str,
+                      self.task_q, self.res_q),
+                daemon=True,
+            )
+            for _ in range(self.workers)
+        ]
+        for p in self.procs:
+            p.start()
+
+    def any_dead(self) -> List[int]:
+        return [p.pid for p in self.procs if not p.is_alive()]
+
+    def close(self) -> None:
+        for _ in self.procs:
+            try:
+                self.task_q.put_nowait(None)
+            except Exception:
+                pass
+        for p in self.procs:
+            p.join(timeout=2.0)
+        for p in self.procs:
+            if p.is_alive():
+                p.terminate()
+                p.join(timeout=2.0)
+        for q_ in (self.task_q, self.res_q):
+            try:
+                q_.close()
+                q_.cancel_join_thread()
+            except Exception:
+                pass
+        try:
+            self.shm.close()
+            self.shm.unlink()
+        except Exception:
+            pass
 
 
 class _H2DStager:
@@ -102,6 +206,7 @@ class ShardedParquetLoader:
         label_column: str = "label_idx",
         transform: Optional[Callable] = None,
         prefetch: int = 2,
+        pool: str = "auto",  # "process" | "thread" | "auto"
     ):
         self.path = str(dataset_path)
         self.files = sorted(str(p) for p in Path(self.path).glob("*.parquet")) or [self.path]
@@ -119,6 +224,9 @@ class ShardedParquetLoader:
             lambda c: preprocess_bytes(c, self.img_height, self.img_width)
         )
         self.prefetch = prefetch
+        if pool not in ("auto", "process", "thread"):
+            raise ValueError(f"pool must be auto|process|thread, got {pool!r}")
+        self.pool = pool
         # (file_idx, row_group_idx) pairs across all files
         self._rg_index: List[Tuple[int, int]] = []
         self._num_rows = 0
@@ -184,29 +292,119 @@ class ShardedParquetLoader:
         finally:
             pool.shutdown(wait=False)
 
-    def __iter__(self):
-        if self.device is None or self.device.type == "cpu":
-            yield from self._batches_cpu()
+    def _batches_proc(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor, Callable]]:
+        """Decode batches via the process pool. Yields
+        ``(images_view, labels, release)`` where ``images_view`` is a
+        zero-copy torch view over a shared-memory slot: the consumer must
+        copy it out (to pinned / to an owned tensor) and then call
+        ``release()`` to return the slot to the ring. Batch order is
+        deterministic (seq-reordered) regardless of worker scheduling."""
+        rows = self._iter_rows()
+        first = next(rows, None)
+        if first is None:
             return
-        # GPU path: background decode thread + pinned double-buffer staging
+        sample = self.transform(first[0])
+        if isinstance(sample, torch.Tensor):
+            sample = sample.numpy()
+        sample = np.asarray(sample)
+        rows = itertools.chain([first], rows)
+
+        pool = _ProcDecodePool(
+            self.transform, sample.shape, sample.dtype, self.batch_size, self.workers)
+        free_slots: deque = deque(range(pool.slots))
+        pending_labels = {}
+        heap: list = []
+        submit_seq = 0
+        yield_seq = 0
+        exhausted = False
+        try:
+            while True:
+                while free_slots and not exhausted:
+                    chunk = list(itertools.islice(rows, self.batch_size))
+                    if not chunk:
+                        exhausted = True
+                        break
+                    slot = free_slots.popleft()
+                    pool.task_q.put((submit_seq, slot, [c for c, _ in chunk]))
+                    pending_labels[submit_seq] = torch.tensor(
+                        [l for _, l in chunk], dtype=torch.long)
+                    submit_seq += 1
+                    if len(chunk) < self.batch_size:
+                        exhausted = True
+                if exhausted and yield_seq == submit_seq:
+                    return
+                while not heap or heap[0][0] != yield_seq:
+                    try:
+                        status, seq, slot, payload = pool.res_q.get(timeout=5.0)
+                    except queue.Empty:
+                        dead = pool.any_dead()
+                        if dead:
+                            raise RuntimeError(
+                                f"decode worker(s) died (pids {dead})")
+                        continue
+                    if status == "err":
+                        raise RuntimeError(f"decode failed in worker: {payload}")
+                    heapq.heappush(heap, (seq, slot, payload))
+                seq, slot, n = heapq.heappop(heap)
+                labels = pending_labels.pop(seq)
+                imgs = torch.from_numpy(pool.view[slot, :n])
+                # deque.append is atomic -> safe to call from a consumer thread
+                yield imgs, labels, (lambda s=slot: free_slots.append(s))
+                yield_seq += 1
+        finally:
+            pool.close()
+
+    def _resolve_pool(self) -> str:
+        if self.pool != "auto":
+            return self.pool
+        import multiprocessing as mp
+
+        if mp.current_process().daemon:
+            return "thread"  # daemonic processes cannot fork children
+        # processes pay off where decode must keep up with a GPU; the CPU
+        # path keeps the cheap thread pool (tests, small oracle runs)
+        return "process" if (self.device is not None and self.device.type != "cpu") else "thread"
+
+    def __iter__(self):
+        mode = self._resolve_pool()
+        if self.device is None or self.device.type == "cpu":
+            if mode == "process":
+                for imgs, labels, release in self._batches_proc():
+                    out = imgs.clone()
+                    release()
+                    yield out, labels
+            else:
+                yield from self._batches_cpu()
+            return
+        # GPU path: background decode (process pool or thread pool) + pinned
+        # double-buffer staging on a side stream
         if self._stager is None:
             self._stager = _H2DStager(self.device, depth=max(2, self.prefetch))
         q: "queue.Queue" = queue.Queue(maxsize=self.prefetch)
         stop = threading.Event()
 
+        def _bounded_put(item) -> bool:
+            # bounded put so the thread can exit promptly once the
+            # consumer is gone (avoids a blocked thread at teardown)
+            while not stop.is_set():
+                try:
+                    q.put(item, timeout=0.25)
+                    return True
+                except queue.Full:
+                    continue
+            return False
+
         def producer():
             try:
-                for batch in self._batches_cpu():
-                    # bounded put so the thread can exit promptly once the
-                    # consumer is gone (avoids a blocked thread at teardown)
-                    while not stop.is_set():
-                        try:
-                            q.put(batch, timeout=0.25)
-                            break
-                        except queue.Full:
-                            continue
-                    if stop.is_set():
-                        return
+                if mode == "process":
+                    for imgs, labels, release in self._batches_proc():
+                        if not _bounded_put((imgs, labels, release)):
+                            release()
+                            return
+                else:
+                    for batch in self._batches_cpu():
+                        if not _bounded_put((batch[0], batch[1], None)):
+                            return
             finally:
                 try:
                     q.put_nowait(None)
@@ -220,8 +418,18 @@ class ShardedParquetLoader:
                 item = q.get()
                 if item is None:
                     return
-                di, dl, ev = self._stager.stage(*item)
-                torch.cuda.current_stream(self.device).wait_event(ev)
+                imgs, labels, release = item
+                di, dl, ev = self._stager.stage(imgs, labels)
+                if release is not None:
+                    release()  # stage() memcpy'd the slot into pinned
+                cs = torch.cuda.current_stream(self.device)
+                cs.wait_event(ev)
+                # the device tensors were allocated on the side stream; tell
+                # the caching allocator they are consumed on the compute
+                # stream, or a freed block could be handed to a later
+                # side-stream H2D copy while compute still reads it
+                di.record_stream(cs)
+                dl.record_stream(cs)
                 yield di, dl
         finally:
             stop.set()
@@ -229,10 +437,12 @@ class ShardedParquetLoader:
             # left inside native code at interpreter teardown
             while not q.empty():
                 try:
-                    q.get_nowait()
+                    item = q.get_nowait()
+                    if item is not None and item[2] is not None:
+                        item[2]()
                 except queue.Empty:
                     break
-            t.join(timeout=5.0)
+            t.join(timeout=10.0)
 
 
 class Converter:

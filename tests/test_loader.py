@@ -119,3 +119,59 @@ def test_loader_row_level_fallback_more_ranks_than_groups(tmp_path):
         assert labels, f"rank {r} got an empty shard"
         seen.extend(labels)
     assert sorted(seen) == list(range(10))  # disjoint + exhaustive
+
+
+def test_loader_process_pool_matches_thread_pool(ddlw_home):
+    """The forked decode pool must deliver byte-identical batches in the
+    same deterministic order as the thread path (seq reorder buffer)."""
+    tbl = _make_table(40, seed=9)
+    conv = make_converter(tbl, row_group_rows=8)
+    kw = dict(batch_size=8, cur_shard=0, shard_count=1, num_epochs=1,
+              img_height=16, img_width=16)
+    with conv.make_torch_dataset(pool="thread", **kw) as loader:
+        thread_batches = [(i.clone(), l.clone()) for i, l in loader]
+    with conv.make_torch_dataset(pool="process", workers_count=3, **kw) as loader:
+        proc_batches = [(i.clone(), l.clone()) for i, l in loader]
+    assert len(proc_batches) == len(thread_batches)
+    for (ti, tl), (pi, pl) in zip(thread_batches, proc_batches):
+        assert torch.equal(ti, pi)
+        assert torch.equal(tl, pl)
+    conv.delete()
+
+
+def test_loader_process_pool_infinite_cycling(ddlw_home):
+    tbl = _make_table(16, seed=4)
+    conv = make_converter(tbl, row_group_rows=8)
+    with conv.make_torch_dataset(
+        batch_size=8, num_epochs=None, img_height=16, img_width=16,
+        pool="process", workers_count=2,
+    ) as loader:
+        it = iter(loader)
+        batches = [next(it) for _ in range(5)]  # > 2 epochs worth
+        it.close()
+    assert all(b[0].shape[0] == 8 for b in batches)
+    conv.delete()
+
+
+def test_loader_process_pool_surfaces_decode_error(ddlw_home):
+    """A transform raising in a worker must raise in the consumer, not hang."""
+    tbl = _make_table(8, seed=1)
+    conv = make_converter(tbl, row_group_rows=8)
+
+    with conv.make_torch_dataset(
+        batch_size=4, num_epochs=1, pool="process", workers_count=2,
+        transform=_boom,
+    ) as loader:
+        with pytest.raises(RuntimeError, match="decode failed"):
+            list(loader)
+    conv.delete()
+
+
+def _boom(content):
+    # raise only inside the forked decode workers (the main-process shape
+    # probe must pass so the pool actually starts)
+    import multiprocessing as mp
+
+    if mp.current_process().daemon:
+        raise ValueError("bad jpeg")
+    return np.zeros((4, 4, 3), np.uint8)
