@@ -147,6 +147,23 @@ def load_model(model_uri: str) -> PyFuncModel:
 
 def _pool_worker(model_uri: str, env: Dict[str, str], task_q, result_q) -> None:
     os.environ.update(env)
+    # Workers are non-daemonic (a packaged model may fork its own decode
+    # pool — daemonic processes cannot have children), so guard against
+    # outliving a crashed parent: poll the parent pid and exit if it changes
+    # (orphaned -> reparented).
+    import threading as _t
+
+    ppid = os.getppid()
+
+    def _watchdog():
+        import time as _time
+
+        while True:
+            _time.sleep(2.0)
+            if os.getppid() != ppid:
+                os._exit(0)
+
+    _t.Thread(target=_watchdog, daemon=True).start()
     try:
         m = load_model(model_uri)
         result_q.put(("ready", os.getpid(), None))
@@ -193,15 +210,24 @@ class PredictUDF:
         self._result_q = ctx.Queue()
         self._task_qs = []
         self._procs = []
+        self._closed = False
+        # non-daemon workers would block interpreter exit (multiprocessing
+        # joins children at shutdown) if the caller forgets close(); our
+        # atexit hook runs before mp's join and sends the sentinels
+        import atexit
+
+        atexit.register(self.close)
         base_env = {"DDLW_TRACKING_URI": tracking.get_tracking_uri()}
         for i in range(num_workers):
             env = dict(base_env)
             if gpus:
                 env["HIP_VISIBLE_DEVICES"] = str(gpus[i % len(gpus)])
             tq = ctx.Queue()
+            # daemon=False so the worker can fork a decode pool (see
+            # _pool_worker's orphan watchdog for the cleanup guarantee)
             p = ctx.Process(
                 target=_pool_worker, args=(model_uri, env, tq, self._result_q),
-                daemon=True,
+                daemon=False,
             )
             p.start()
             self._task_qs.append(tq)
@@ -278,6 +304,9 @@ class PredictUDF:
         return out
 
     def close(self) -> None:
+        if self._closed:
+            return
+        self._closed = True
         for tq in self._task_qs:
             try:
                 tq.put(None)

@@ -132,8 +132,8 @@ def main():
     print("single-node:", list(preds10))
 
     # distributed fan-out on up to 1000 rows (P2/03:466-472)
-    udf = predict_udf(uri, num_workers=args.workers)
-    preds = udf(rows[:1000])
+    with predict_udf(uri, num_workers=args.workers) as udf:
+        preds = udf(rows[:1000])
     print(f"fanned out {len(preds)} predictions; first 5: {preds[:5]}")
     assert list(preds10) == [str(p) for p in preds[:10]]
 

@@ -74,8 +74,8 @@ def test_predict_udf_matches_single_node(ddlw_home):
     uri = _package(ddlw_home)
     contents, _ = make_synthetic_dataset(10, 16, 16, num_classes=5, seed=2, jpeg=True)
     single = list(load_model(uri).predict(contents))
-    udf = predict_udf(uri, num_workers=3, gpus=[])
-    fanned = udf(contents)
+    with predict_udf(uri, num_workers=3, gpus=[]) as udf:
+        fanned = udf(contents)
     assert fanned == [str(s) for s in single]
 
 
@@ -119,3 +119,49 @@ def test_predict_udf_uneven_split_no_empty_parts(ddlw_home):
         assert fanned == [str(s) for s in single]
         # reuse the pool: a second call must be unaffected
         assert udf(contents[:2]) == [str(s) for s in single[:2]]
+
+
+def _times_two(content):
+    import numpy as np
+
+    return np.frombuffer(content, dtype=np.uint8).reshape(2, 2) * 2
+
+
+def test_parallel_decoder_matches_serial():
+    from ddlw_amd.data.decode import ParallelDecoder
+
+    rows = [bytes([i, i + 1, i + 2, i + 3]) for i in range(0, 40, 4)]
+    with ParallelDecoder(_times_two, workers=3, chunk_size=3) as dec:
+        out = dec.map(rows)
+        # pool reuse across calls
+        out2 = dec.map(rows[:4])
+    serial = ParallelDecoder(_times_two, workers=0).map(rows)
+    assert torch.equal(out, serial)
+    assert torch.equal(out2, serial[:4])
+
+
+class _ForkingModel(PythonModel):
+    """Model whose predict forks a decode pool — requires non-daemonic
+    UDF workers (the inference decode path, VERDICT r1 #4)."""
+
+    def load_context(self, context):
+        pass
+
+    def predict(self, context, rows):
+        from ddlw_amd.data.decode import ParallelDecoder
+
+        with ParallelDecoder(_times_two, workers=2, chunk_size=2) as dec:
+            out = dec.map(rows)
+        return [str(int(t.sum())) for t in out]
+
+
+def test_predict_udf_worker_can_fork_decode_pool(ddlw_home):
+    from ddlw_amd.infer.pyfunc import log_model, predict_udf
+
+    tracking.set_experiment("udf_fork")
+    with tracking.start_run():
+        uri = log_model("forking", _ForkingModel())
+    rows = [bytes([i, i + 1, i + 2, i + 3]) for i in range(0, 24, 4)]
+    expect = [str(int(_times_two(r).sum())) for r in rows]
+    with predict_udf(uri, num_workers=2, gpus=[]) as udf:
+        assert udf(rows) == expect
