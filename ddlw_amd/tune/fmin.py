@@ -177,8 +177,12 @@ class _TPE:
         return out
 
     def suggest(self, space: Dict[str, Any], history: List[dict], rng) -> Dict[str, Any]:
-        """Return one new internal-parameter dict."""
+        """Return one new internal-parameter dict. Trials with an empty
+        result are PENDING (e.g. same-batch picks not yet run): they join
+        the "bad" density so concurrent suggestions spread out instead of
+        clustering (hyperopt's pending-trial treatment)."""
         done = [t for t in history if t["result"].get("status") == STATUS_OK]
+        pending = [t for t in history if not t["result"]]
         if len(done) < self.n_startup_jobs:
             return {k: sample_param(spec, rng) for k, spec in space.items()}
         losses = np.array([t["result"]["loss"] for t in done])
@@ -190,7 +194,7 @@ class _TPE:
         out: Dict[str, Any] = {}
         for key, spec in space.items():
             gv = np.array([t["params"][key] for t in good], dtype=float)
-            bv = np.array([t["params"][key] for t in bad], dtype=float)
+            bv = np.array([t["params"][key] for t in bad + pending], dtype=float)
             if spec.kind == "choice":
                 k = len(spec.options)
                 gc = np.bincount(gv.astype(int), minlength=k) + 1.0
@@ -255,14 +259,13 @@ def fmin(
     while len(trials.trials) < max_evals:
         batch_n = min(trials.parallelism, max_evals - len(trials.trials))
         internal_batch = []
+        pending = []  # same-batch picks, passed to TPE as result-less trials
         for _ in range(batch_n):
             if algo_name == "rand":
                 internal = {k: sample_param(s, rng) for k, s in space.items()}
             else:
-                internal = suggester.suggest(space, trials.trials + [
-                    # treat same-batch picks as pending (no result) — they
-                    # don't enter the TPE split but avoid duplicate sampling
-                ], rng)
+                internal = suggester.suggest(space, trials.trials + pending, rng)
+                pending.append({"tid": -1, "params": internal, "result": {}})
             internal_batch.append(internal)
         external_batch = [
             {k: externalize(space[k], v) for k, v in internal.items()}
