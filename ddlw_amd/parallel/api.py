@@ -137,42 +137,73 @@ def allreduce_metrics(metrics: Dict[str, float]) -> Dict[str, float]:
     return {k: v for k, v in zip(keys, t.cpu().tolist())}
 
 
-def broadcast_parameters(module_or_params, root_rank: int = 0) -> None:
+def broadcast_parameters(module_or_params, root_rank: int = 0,
+                         bucket_cap_mb: float = 64.0) -> None:
     """Broadcast model parameters (+ buffers) from root (C3). Ensures every
     rank starts from rank-0's init or restored checkpoint
-    (reference comment ``.../03_model_training_distributed.py:305-307``)."""
+    (reference comment ``.../03_model_training_distributed.py:305-307``).
+
+    Tensors are coalesced into per-(dtype, device) flat buckets — one
+    collective per bucket (~#dtypes total) instead of one per tensor
+    (ResNet-50 has ~320 state tensors; Horovod's fusion exists for exactly
+    this). CPU-resident buffers ride a single staged device round-trip per
+    bucket when the backend is RCCL."""
     if not _pg_active():
         return
     if isinstance(module_or_params, torch.nn.Module):
         tensors: Iterable[torch.Tensor] = list(module_or_params.state_dict().values())
     else:
         tensors = list(module_or_params)
+    tensors = [
+        t for t in tensors
+        if t.dtype.is_floating_point or t.dtype in (torch.int64, torch.int32, torch.uint8)
+    ]
     use_nccl = dist.get_backend() == "nccl"
+    cap = int(bucket_cap_mb * 1024 * 1024)
+    groups: Dict[Tuple[torch.dtype, str], List[torch.Tensor]] = {}
     for t in tensors:
-        if t.dtype.is_floating_point or t.dtype in (torch.int64, torch.int32, torch.uint8):
-            if _native_comm is not None and t.is_cuda and t.is_contiguous():
-                _native_comm.broadcast_(t, root=root_rank)
-            elif use_nccl and not t.is_cuda:
-                # CPU-resident buffers (e.g. BN step counters) ride a device
-                # round-trip — NCCL/RCCL moves device tensors only
-                d = t.to(torch.device("cuda", torch.cuda.current_device()))
+        groups.setdefault((t.dtype, t.device.type), []).append(t)
+    for (dtype, devtype), ts in groups.items():
+        start = 0
+        while start < len(ts):
+            nbytes, end = 0, start
+            while end < len(ts) and (
+                end == start
+                or nbytes + ts[end].numel() * ts[end].element_size() <= cap
+            ):
+                nbytes += ts[end].numel() * ts[end].element_size()
+                end += 1
+            chunk = ts[start:end]
+            start = end
+            flat = torch._utils._flatten_dense_tensors([c.detach() for c in chunk])
+            if _native_comm is not None and flat.is_cuda:
+                _native_comm.broadcast_(flat, root=root_rank)
+                torch.cuda.synchronize()
+            elif use_nccl and devtype == "cpu":
+                # RCCL moves device tensors only: one staged round-trip
+                d = flat.to(torch.device("cuda", torch.cuda.current_device()))
                 dist.broadcast(d, src=root_rank)
-                t.copy_(d.cpu())
+                flat = d.cpu()
             else:
-                dist.broadcast(t, src=root_rank)
-    if _native_comm is not None:
-        torch.cuda.synchronize()
+                dist.broadcast(flat, src=root_rank)
+            with torch.no_grad():
+                for t, piece in zip(
+                    chunk, torch._utils._unflatten_dense_tensors(flat, chunk)
+                ):
+                    t.copy_(piece)  # identity on root
 
 
 def broadcast_optimizer_state(optimizer: torch.optim.Optimizer, root_rank: int = 0) -> None:
     if not _pg_active():
         return
-    for group in optimizer.param_groups:
-        for p in group["params"]:
-            state = optimizer.state.get(p, {})
-            for v in state.values():
-                if torch.is_tensor(v):
-                    dist.broadcast(v, src=root_rank)
+    tensors = [
+        v
+        for group in optimizer.param_groups
+        for p in group["params"]
+        for v in optimizer.state.get(p, {}).values()
+        if torch.is_tensor(v)
+    ]
+    broadcast_parameters(tensors, root_rank=root_rank)
 
 
 # --------------------------------------------------------------------------- #

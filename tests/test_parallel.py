@@ -217,3 +217,78 @@ def test_distributed_optimizer_with_unused_param():
     out = Runner(np=2, timeout_s=120).run(_unused_param_worker, seed=3)
     assert out["unused_grad_none"]
     assert torch.isfinite(out["w"]).all()
+
+
+def _coalesced_broadcast_worker(seed):
+    """broadcast must coalesce into per-(dtype,device) flat
+    buckets — collective count ~ #dtype groups, not #tensors — and still
+    leave every rank with rank-0's exact state."""
+    import torch.distributed as dist
+
+    from ddlw_amd.parallel import api
+
+    torch.manual_seed(seed + api.rank())  # divergent init across ranks
+    m = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 8, 3), torch.nn.BatchNorm2d(8), torch.nn.Linear(8, 4)
+    )
+    m[2].weight.data = m[2].weight.data.to(torch.bfloat16)  # second dtype
+    n_tensors = len(list(m.state_dict().values()))
+    calls = []
+    orig = dist.broadcast
+
+    def counting(t, *a, **k):
+        calls.append(t.numel())
+        return orig(t, *a, **k)
+
+    dist.broadcast = counting
+    try:
+        api.broadcast_parameters(m, root_rank=0)
+    finally:
+        dist.broadcast = orig
+    assert len(calls) <= 4, f"{len(calls)} collectives for {n_tensors} tensors"
+    assert len(calls) < n_tensors
+    import hashlib
+
+    h = hashlib.sha256()
+    for k, v in m.state_dict().items():
+        h.update(k.encode())
+        h.update(v.float().numpy().tobytes())
+    return h.hexdigest()
+
+
+def test_broadcast_parameters_coalesced():
+    # run twice with different seeds; rank-0 digest returned — both ranks
+    # asserted equal inside via the returned digest equality across runs
+    d1 = Runner(np=2, timeout_s=120).run(_coalesced_broadcast_worker, seed=10)
+    d2 = Runner(np=2, timeout_s=120).run(_coalesced_broadcast_worker, seed=10)
+    assert d1 == d2  # deterministic given rank-0 seed
+
+
+def _coalesced_broadcast_parity_worker(seed):
+    """All ranks must hold identical state after the coalesced broadcast."""
+    from ddlw_amd.parallel import api
+
+    torch.manual_seed(seed + api.rank())
+    m = torch.nn.Sequential(torch.nn.Linear(6, 6), torch.nn.BatchNorm1d(6))
+    m[0].weight.data = m[0].weight.data.to(torch.bfloat16)
+    api.broadcast_parameters(m, root_rank=0)
+    sd = {k: v.clone() for k, v in m.state_dict().items()}
+    # gather every rank's digest on rank 0 via all_reduce of a checksum
+    import hashlib
+
+    h = hashlib.sha256()
+    for k in sorted(sd):
+        h.update(sd[k].float().numpy().tobytes())
+    digest = int.from_bytes(h.digest()[:6], "big")
+    t = torch.tensor([float(digest)], dtype=torch.float64)
+    mx, mn = t.clone(), t.clone()
+    import torch.distributed as dist
+
+    dist.all_reduce(mx, op=dist.ReduceOp.MAX)
+    dist.all_reduce(mn, op=dist.ReduceOp.MIN)
+    assert torch.equal(mx, mn), "ranks diverge after broadcast"
+    return True
+
+
+def test_broadcast_parameters_rank_parity():
+    assert Runner(np=2, timeout_s=120).run(_coalesced_broadcast_parity_worker, seed=3)
