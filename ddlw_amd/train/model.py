@@ -157,6 +157,16 @@ class Model:
     ) -> History:
         if self.optimizer is None:
             raise RuntimeError("call compile() before fit()")
+        # An infinite stream (Petastorm num_epochs=None semantics) with no
+        # steps_per_epoch would never end an epoch — fail loudly instead
+        # (Keras shares this footgun; we don't).
+        if steps_per_epoch is None and getattr(data, "num_epochs", 0) is None:
+            raise ValueError(
+                "fit(): dataset advertises infinite epochs (num_epochs=None) "
+                "but steps_per_epoch was not given — the first epoch would "
+                "never end. Pass steps_per_epoch (e.g. len(converter) // "
+                "(batch_size * size()))."
+            )
         from ..parallel import api
 
         self.stop_training = False

@@ -203,3 +203,22 @@ def test_fused_optimizer_fresh_instance_resume_bf16():
 
         assert torch.allclose(a32, c32, atol=1e-6), cls.__name__
         assert torch.equal(a16, c16), cls.__name__
+
+
+def test_fit_rejects_infinite_loader_without_steps(ddlw_home):
+    """an infinite stream + steps_per_epoch=None must
+    raise, not spin forever."""
+    import pyarrow as pa
+    import pytest
+
+    from ddlw_amd.data import make_converter, make_synthetic_dataset
+    from ddlw_amd.train import Model
+
+    contents, labels = make_synthetic_dataset(8, 16, 16, num_classes=3, jpeg=True)
+    tbl = pa.table({"content": pa.array(contents, pa.binary()), "label_idx": labels})
+    conv = make_converter(tbl, row_group_rows=8)
+    m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.05)
+    with conv.make_torch_dataset(batch_size=4, num_epochs=None, img_height=16, img_width=16) as ds:
+        with pytest.raises(ValueError, match="infinite"):
+            m.fit(ds, epochs=1, verbose=0)
+    conv.delete()
