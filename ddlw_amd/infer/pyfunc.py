@@ -332,3 +332,48 @@ def predict_udf(
     """Return a callable ``udf(rows) -> list[str]`` backed by a persistent
     worker pool; call ``.close()`` (or use as a context manager) when done."""
     return PredictUDF(model_uri, num_workers=num_workers, gpus=gpus)
+
+
+def predict_table(
+    model_uri: str,
+    source,
+    content_column: str = "content",
+    output_column: str = "prediction",
+    num_workers: Optional[int] = None,
+    gpus: Optional[List[int]] = None,
+    output_path: Optional[str] = None,
+    limit: Optional[int] = None,
+):
+    """Table-in -> table-out batch inference: the reference's
+    ``df.withColumn('prediction', loaded_model_udf('content'))`` contract
+    (``Part 2 .../03_pyfunc_distributed_inference.py:466-472``).
+
+    ``source`` is a pyarrow Table or a Parquet file/dataset-dir path. The
+    ``content_column`` rows are fanned across the worker pool; the returned
+    Table is the input (optionally ``limit``-ed, like the reference's
+    ``.limit(1000)``) plus a string ``output_column``. When ``output_path``
+    is given the result is also written there as Parquet.
+    """
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    if isinstance(source, (str, Path)):
+        p = Path(source)
+        if p.is_dir():
+            parts = sorted(p.glob("*.parquet"))
+            table = pa.concat_tables([pq.read_table(f) for f in parts])
+        else:
+            table = pq.read_table(p)
+    else:
+        table = source
+    if limit is not None:
+        table = table.slice(0, limit)
+    rows = table.column(content_column).to_pylist()
+    with PredictUDF(model_uri, num_workers=num_workers, gpus=gpus) as udf:
+        preds = udf(rows)
+    out = table.append_column(output_column, pa.array(preds, pa.string()))
+    if output_path is not None:
+        out_p = Path(output_path)
+        out_p.parent.mkdir(parents=True, exist_ok=True)
+        pq.write_table(out, out_p)
+    return out

@@ -165,3 +165,25 @@ def test_predict_udf_worker_can_fork_decode_pool(ddlw_home):
     expect = [str(int(_times_two(r).sum())) for r in rows]
     with predict_udf(uri, num_workers=2, gpus=[]) as udf:
         assert udf(rows) == expect
+
+
+def test_predict_table_adds_prediction_column(ddlw_home, tmp_path):
+    """Table-in/table-out contract (reference P2/03:466-472): input table +
+    string prediction column; parquet round-trip; limit like .limit(1000)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from ddlw_amd.infer import load_model, predict_table
+
+    uri = _package(ddlw_home)
+    contents, labels = make_synthetic_dataset(12, 16, 16, num_classes=5, seed=7, jpeg=True)
+    tbl = pa.table({"content": pa.array(contents, pa.binary()),
+                    "label_idx": labels})
+    out_pq = tmp_path / "pred" / "out.parquet"
+    out = predict_table(uri, tbl, num_workers=2, gpus=[], limit=10,
+                        output_path=str(out_pq))
+    assert out.num_rows == 10
+    assert out.column_names == ["content", "label_idx", "prediction"]
+    single = [str(s) for s in load_model(uri).predict(contents[:10])]
+    assert out.column("prediction").to_pylist() == single
+    assert pq.read_table(out_pq).column("prediction").to_pylist() == single
