@@ -21,6 +21,13 @@ waits on the outstanding works before applying the update.
 
 Degenerate world (size 1 / no env rendezvous) runs without a process group —
 the reference's ``HorovodRunner(np=-1)`` in-process smoke mode (:385-394).
+
+Data-plane decision (round 2): torch's ProcessGroupNCCL — which IS RCCL on
+ROCm — is the one multi-GPU data plane. A round-1 ctypes RCCL wrapper was
+removed: it duplicated the same librccl calls behind a layer that could
+never be validated multi-rank on the 1-GPU dev lease, and the parts that
+matter for xGMI performance (bucket fusion, launch-during-backward overlap,
+bucket sizing for the 7-link mesh) live HERE, above the collective call.
 """
 from __future__ import annotations
 
@@ -34,7 +41,6 @@ _initialized = False
 _world_size = 1
 _rank = 0
 _local_rank = 0
-_native_comm = None  # NativeComm when DDLW_NATIVE_CC=1 (see parallel.native)
 
 
 def init(backend: Optional[str] = None, timeout_s: float = 300.0) -> None:
@@ -65,24 +71,11 @@ def init(backend: Optional[str] = None, timeout_s: float = 300.0) -> None:
             backend=backend,
             timeout=datetime.timedelta(seconds=timeout_s),
         )
-        if backend == "nccl":
-            from . import native as _native_mod
-
-            if _native_mod.enabled():
-                # opt-in native RCCL data plane (DDLW_NATIVE_CC=1): gradient
-                # buckets + broadcasts go through libddlw_rccl on a side
-                # stream instead of ProcessGroupNCCL (both ARE RCCL/xGMI)
-                global _native_comm
-                device = torch.device("cuda", torch.cuda.current_device())
-                _native_comm = _native_mod.NativeComm(world, _rank, device)
     _initialized = True
 
 
 def shutdown() -> None:
-    global _initialized, _world_size, _rank, _local_rank, _native_comm
-    if _native_comm is not None:
-        _native_comm.destroy()
-        _native_comm = None
+    global _initialized, _world_size, _rank, _local_rank
     if _initialized and dist.is_initialized():
         dist.destroy_process_group()
     _initialized = False
@@ -176,10 +169,7 @@ def broadcast_parameters(module_or_params, root_rank: int = 0,
             chunk = ts[start:end]
             start = end
             flat = torch._utils._flatten_dense_tensors([c.detach() for c in chunk])
-            if _native_comm is not None and flat.is_cuda:
-                _native_comm.broadcast_(flat, root=root_rank)
-                torch.cuda.synchronize()
-            elif use_nccl and devtype == "cpu":
+            if use_nccl and devtype == "cpu":
                 # RCCL moves device tensors only: one staged round-trip
                 d = flat.to(torch.device("cuda", torch.cuda.current_device()))
                 dist.broadcast(d, src=root_rank)
@@ -212,14 +202,13 @@ def broadcast_optimizer_state(optimizer: torch.optim.Optimizer, root_rank: int =
 
 
 class _Bucket:
-    __slots__ = ("params", "bytes", "flat", "work", "ev", "ready", "live")
+    __slots__ = ("params", "bytes", "flat", "work", "ready", "live")
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
         self.bytes = 0
         self.flat: Optional[torch.Tensor] = None
         self.work = None
-        self.ev = None  # native-path completion event
         self.ready = 0
         self.live: List[torch.nn.Parameter] = []  # params with grads this step
 
@@ -300,12 +289,9 @@ class DistributedOptimizer:
                 f"allreduce[{flat.numel() * flat.element_size() >> 20}MiB]",
                 "collective", _t.time() * 1e6, 1.0, tid=_rank,
             )
-        # async: RCCL enqueues on a comm stream (overlaps backward); the
-        # handle/event orders the optimizer stream behind the collective
-        if _native_comm is not None:
-            bucket.ev = _native_comm.allreduce_(flat, average=self.average)
-        else:
-            bucket.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
+        # async: RCCL enqueues on its own comm stream (overlaps backward);
+        # the work handle orders the optimizer behind the collective
+        bucket.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
         bucket.flat = flat
 
     # -- torch optimizer surface ---------------------------------------- #
@@ -344,11 +330,7 @@ class DistributedOptimizer:
             if b.work is not None:
                 b.work.wait()
                 b.work = None
-            if b.ev is not None:
-                # native path: ncclAvg already averaged; just order streams
-                torch.cuda.current_stream().wait_event(b.ev)
-                b.ev = None
-            elif self.average:
+            if self.average:
                 b.flat.div_(_world_size)
             grads = [p.grad for p in b.live]
             for p, g in zip(
