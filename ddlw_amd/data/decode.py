@@ -47,7 +47,7 @@ class ParallelDecoder:
     """
 
     def __init__(self, transform: Callable, workers: Optional[int] = None,
-                 chunk_size: int = 64):
+                 chunk_size: int = 64, pin: bool = False):
         import multiprocessing as mp
 
         self.transform = transform
@@ -57,13 +57,14 @@ class ParallelDecoder:
         if mp.current_process().daemon:
             workers = 0  # cannot fork children
         self.workers = workers
+        self.pin = pin
         self._pool: Optional[_ProcDecodePool] = None
 
     def _ensure_pool(self, sample: np.ndarray) -> _ProcDecodePool:
         if self._pool is None:
             self._pool = _ProcDecodePool(
                 self.transform, sample.shape, sample.dtype,
-                self.chunk_size, self.workers)
+                self.chunk_size, self.workers, pin=self.pin)
         return self._pool
 
     def map(self, contents: Sequence) -> torch.Tensor:
@@ -110,6 +111,66 @@ class ParallelDecoder:
             free.append(slot)
             done += 1
         return torch.from_numpy(out)
+
+    def imap(self, contents: Sequence):
+        """Pipelined chunk iterator: yields decoded ``chunk_size`` batches
+        IN ORDER while later chunks decode concurrently in the pool — so a
+        GPU consumer overlaps forward with decode. The yielded tensor is a
+        zero-copy view of a pool slot and is valid ONLY until the next
+        ``next()`` (the slot is recycled then); finish any device copy
+        within the iteration. When the pool is pinned the view is DMA-able
+        (``.cuda(non_blocking=True)`` is a true async copy)."""
+        n = len(contents)
+        if n == 0:
+            return
+        first = self.transform(contents[0])
+        if isinstance(first, torch.Tensor):
+            first = first.numpy()
+        first = np.asarray(first)
+        chunks = [(s, min(s + self.chunk_size, n))
+                  for s in range(0, n, self.chunk_size)]
+        if self.workers <= 1:
+            for s, e in chunks:
+                out = np.empty((e - s,) + first.shape, dtype=first.dtype)
+                for i in range(s, e):
+                    r = self.transform(contents[i])
+                    out[i - s] = r.numpy() if isinstance(r, torch.Tensor) else r
+                yield torch.from_numpy(out)
+            return
+        pool = self._ensure_pool(first)
+        free = list(range(pool.slots))
+        ready = {}  # seq -> slot
+        next_submit = 0
+        next_yield = 0
+        prev_slot = None
+        try:
+            while next_yield < len(chunks):
+                while free and next_submit < len(chunks):
+                    slot = free.pop()
+                    s, e = chunks[next_submit]
+                    pool.task_q.put((next_submit, slot, list(contents[s:e])))
+                    next_submit += 1
+                while next_yield not in ready:
+                    try:
+                        status, seq, slot, payload = pool.res_q.get(timeout=10.0)
+                    except queue.Empty:
+                        dead = pool.any_dead()
+                        if dead:
+                            raise RuntimeError(f"decode worker(s) died (pids {dead})")
+                        continue
+                    if status == "err":
+                        raise RuntimeError(f"decode failed in worker: {payload}")
+                    ready[seq] = (slot, payload)
+                slot, cnt = ready.pop(next_yield)
+                view = torch.from_numpy(pool.view[slot, :cnt])
+                if prev_slot is not None:
+                    free.append(prev_slot)  # released one step late: the
+                prev_slot = slot            # consumer just finished with it
+                yield view
+                next_yield += 1
+        finally:
+            if prev_slot is not None:
+                free.append(prev_slot)
 
     def close(self) -> None:
         if self._pool is not None:

@@ -75,6 +75,18 @@ def _decode_worker(transform, shm_name, view_shape, dtype_str, task_q, res_q):
                 res_q.put(("err", seq, slot, repr(e)))
     finally:
         shm.close()
+        # flush the result queue's feeder thread, then hard-exit: a forked
+        # child of a HIP-initialized parent must NOT run the inherited HIP
+        # atexit handlers (they touch runtime state owned by the parent and
+        # can crash the child at teardown)
+        try:
+            res_q.close()
+            res_q.join_thread()
+        except Exception:
+            pass
+        import os as _os
+
+        _os._exit(0)
 
 
 class _ProcDecodePool:
@@ -92,7 +104,7 @@ class _ProcDecodePool:
     SHM_CAP_BYTES = 8 << 30  # ring cap; also caps in-flight parallelism
 
     def __init__(self, transform, sample_shape, sample_dtype, batch_size: int,
-                 workers: int):
+                 workers: int, pin: bool = False):
         import multiprocessing as mp
         from multiprocessing import shared_memory
 
@@ -108,6 +120,18 @@ class _ProcDecodePool:
             create=True, size=max(slot_bytes * self.slots, 1))
         view_shape = (self.slots,) + self.slot_shape
         self.view = np.ndarray(view_shape, dtype=self.np_dtype, buffer=self.shm.buf)
+        # Pin the ring: the shm pages are MAP_SHARED, i.e. the SAME physical
+        # pages the forked workers write — hipHostRegister-ing the parent's
+        # mapping makes every slot DMA-able, so H2D copies run straight from
+        # the decode output with no staging memcpy.
+        self.pinned = False
+        if pin:
+            try:
+                r = torch.cuda.cudart().cudaHostRegister(
+                    int(self.view.ctypes.data), int(self.shm.size), 0)
+                self.pinned = (int(r) == 0)
+            except Exception:
+                self.pinned = False
         self.task_q = ctx.Queue()
         self.res_q = ctx.Queue()
         self.procs = [
@@ -143,6 +167,12 @@ class _ProcDecodePool:
                 q_.cancel_join_thread()
             except Exception:
                 pass
+        if self.pinned:
+            try:
+                torch.cuda.cudart().cudaHostUnregister(int(self.view.ctypes.data))
+            except Exception:
+                pass
+            self.pinned = False
         try:
             self.shm.close()
             self.shm.unlink()
@@ -310,7 +340,10 @@ class ShardedParquetLoader:
         rows = itertools.chain([first], rows)
 
         pool = _ProcDecodePool(
-            self.transform, sample.shape, sample.dtype, self.batch_size, self.workers)
+            self.transform, sample.shape, sample.dtype, self.batch_size,
+            self.workers,
+            pin=self.device is not None and self.device.type == "cuda")
+        self._pool_info = (pool.slots, pool.pinned)
         free_slots: deque = deque(range(pool.slots))
         pending_labels = {}
         heap: list = []
@@ -413,15 +446,34 @@ class ShardedParquetLoader:
 
         t = threading.Thread(target=producer, daemon=True)
         t.start()
+        pending: deque = deque()  # (event, release): slots with H2D in flight
         try:
             while True:
                 item = q.get()
                 if item is None:
                     return
                 imgs, labels, release = item
-                di, dl, ev = self._stager.stage(imgs, labels)
-                if release is not None:
-                    release()  # stage() memcpy'd the slot into pinned
+                pool_slots, pool_pinned = getattr(self, "_pool_info", (0, False))
+                if release is not None and pool_pinned:
+                    # slot ring is hipHostRegister-ed: DMA straight from the
+                    # slot on the side stream; the slot is released once the
+                    # copy's event completes (deferred, non-blocking drain)
+                    ev = torch.cuda.Event()
+                    with torch.cuda.stream(self._stager.stream):
+                        di = imgs.to(self.device, non_blocking=True)
+                        dl = labels.to(self.device, non_blocking=True)
+                        ev.record(self._stager.stream)
+                    pending.append((ev, release))
+                    while pending and pending[0][0].query():
+                        pending.popleft()[1]()
+                    if len(pending) >= max(2, pool_slots - 2):
+                        ev0, rel0 = pending.popleft()
+                        ev0.synchronize()
+                        rel0()
+                else:
+                    di, dl, ev = self._stager.stage(imgs, labels)
+                    if release is not None:
+                        release()  # stage() memcpy'd the slot into pinned
                 cs = torch.cuda.current_stream(self.device)
                 cs.wait_event(ev)
                 # the device tensors were allocated on the side stream; tell
@@ -433,6 +485,12 @@ class ShardedParquetLoader:
                 yield di, dl
         finally:
             stop.set()
+            # in-flight DMA reads the shm ring — finish it before the
+            # producer (and with it the pool / hostUnregister) tears down
+            if pending:
+                torch.cuda.synchronize(self.device)
+                while pending:
+                    pending.popleft()[1]()
             # drain so the producer can exit, then join it so no thread is
             # left inside native code at interpreter teardown
             while not q.empty():

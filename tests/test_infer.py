@@ -187,3 +187,20 @@ def test_predict_table_adds_prediction_column(ddlw_home, tmp_path):
     single = [str(s) for s in load_model(uri).predict(contents[:10])]
     assert out.column("prediction").to_pylist() == single
     assert pq.read_table(out_pq).column("prediction").to_pylist() == single
+
+
+def test_parallel_decoder_imap_order_and_content():
+    from ddlw_amd.data.decode import ParallelDecoder
+
+    rows = [bytes([i, i + 1, i + 2, i + 3]) for i in range(0, 80, 4)]
+    serial = ParallelDecoder(_times_two, workers=0).map(rows)
+    with ParallelDecoder(_times_two, workers=3, chunk_size=4) as dec:
+        got = []
+        for chunk in dec.imap(rows):
+            got.append(chunk.clone())  # view only valid within iteration
+    out = torch.cat(got)
+    assert torch.equal(out, serial)
+    # uneven tail
+    with ParallelDecoder(_times_two, workers=2, chunk_size=3) as dec:
+        sizes = [c.shape[0] for c in dec.imap(rows[:7])]
+    assert sizes == [3, 3, 1]
