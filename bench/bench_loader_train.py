@@ -47,6 +47,9 @@ def main() -> None:
     p.add_argument("--batch-size", type=int, default=256)
     p.add_argument("--rows", type=int, default=4096)
     p.add_argument("--workers", type=int, default=min(os.cpu_count() or 8, 64))
+    p.add_argument("--loader-only", action="store_true",
+                   help="time the input pipeline alone (no model): upper "
+                        "bound on loader-fed throughput")
     args = p.parse_args()
 
     from ddlw_amd.core.config import setup
@@ -86,6 +89,25 @@ def main() -> None:
         prefetch=4,
     ) as loader:
         it = iter(loader)
+
+        if args.loader_only:
+            for _ in range(warmup):
+                next(it)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(steps):
+                next(it)
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            it.close()
+            conv.delete()
+            print(json.dumps({
+                "metric": "loader images/sec (no model)",
+                "value": round(bs * steps / dt, 2),
+                "ms_per_batch": round(dt / steps * 1000, 3),
+                "workers": args.workers,
+            }), flush=True)
+            return
 
         def step_fn():
             imgs, labs = next(it)  # uint8 NHWC on device, staged via side stream
