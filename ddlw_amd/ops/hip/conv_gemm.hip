@@ -25,6 +25,7 @@
 //   - XCD-aware bijective workgroup swizzle (L2 tile locality).
 #include "common.h"
 #include <cstdlib>
+#include <type_traits>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4_v;
@@ -406,6 +407,239 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
 }
 
 // ---------------------------------------------------------------------------
+// Wide-tile pipelined variant: BM=256, 8 waves (512 thr), THREE LDS K-tile
+// buffers, counted-vmcnt + raw-barrier span (cdna guide §5 T3/T4: a glds
+// stays in flight ACROSS the barrier; `__syncthreads()` would emit a
+// vmcnt(0) drain — the ~20% stall of the 2-buffer structure at 1 block/CU),
+// s_setprio(1) around the MFMA cluster (T5: pays on phase-split schedules).
+// Used on long-K-loop shapes (3x3 layers, deep 1x1s) where the 128² 2-buffer
+// kernel measured 540-720 TF/s ≈ 22-29% of peak (profiles/conv_mfma_pmc.md).
+//
+// Per K-tile glds count = (BM + BN) / 64 (one 8-KiB piece per 64 LDS rows,
+// each piece = 1 glds per wave); the gate before computing tile t+1 is
+// s_waitcnt vmcnt(GLDS) — tile t+2's loads keep flying across the barrier.
+// ---------------------------------------------------------------------------
+template <int BM, int BN>
+__global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
+    int N, int H, int W_, int C, int K, int Ho, int Wo,
+    int R, int S, int stride, int pad, int nwg_swz,
+    int oH, int oW, int oS,
+    unsigned long long magic_wo, unsigned shift_wo,
+    unsigned long long magic_ho, unsigned shift_ho,
+    const bf16_t* __restrict__ accp) {
+  constexpr int BK = 64;
+  constexpr int WAVES = 8;            // 4 (M) x 2 (N)
+  constexpr int WM = BM / 4, WN = BN / 2;
+  constexpr int MF = WM / 16, NF = WN / 16;
+  constexpr int AP = BM / 64;         // 1-KiB A pieces per wave (8 rows each)
+  constexpr int BP = BN / 64;
+  constexpr int GLDS = AP + BP;       // glds per wave per K-tile
+  constexpr int BUF = (BM + BN) * BK; // bf16 elements per K-tile buffer
+  __shared__ __attribute__((aligned(16))) bf16_t smem[3 * BUF];
+
+  const long M = (long)N * Ho * Wo;
+  const int tiles_n = (K + BN - 1) / BN;
+
+  int wg = blockIdx.x;
+  {
+    int nwg = nwg_swz;
+    int q = nwg >> 3, rm = nwg & 7;
+    int xcd = wg & 7, i = wg >> 3;
+    wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + i;
+  }
+  const int tile_n = wg % tiles_n;
+  const long tile_m = wg / tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+
+  const int prow = lane >> 3;
+  const int pcol8 = lane & 7;
+  long a_m[AP];
+  int a_hb[AP], a_wb[AP];
+  const bf16_t* a_base[AP];
+  #pragma unroll
+  for (int p = 0; p < AP; ++p) {
+    int row = (wave * AP + p) * 8 + prow;
+    long m = tile_m * BM + row;
+    a_m[p] = m;
+    if (m < M) {
+      unsigned mu = (unsigned)m;
+      unsigned q1 = mdiv(mu, magic_wo, shift_wo);
+      int wo = (int)(mu - q1 * (unsigned)Wo);
+      unsigned n_u = mdiv(q1, magic_ho, shift_ho);
+      int ho = (int)(q1 - n_u * (unsigned)Ho);
+      int n = (int)n_u;
+      a_hb[p] = ho * stride - pad;
+      a_wb[p] = wo * stride - pad;
+      a_base[p] = x + (((long)n * H + a_hb[p]) * W_ + a_wb[p]) * C +
+                  (pcol8 ^ prow) * 8;
+    } else {
+      a_hb[p] = -100000;
+      a_wb[p] = -100000;
+      a_base[p] = zpage;
+    }
+  }
+  const long KRS = (long)R * S * C;
+  const bf16_t* b_base[BP];
+  bool b_ok[BP];
+  #pragma unroll
+  for (int p = 0; p < BP; ++p) {
+    int row = (wave * BP + p) * 8 + prow;
+    int j = tile_n * BN + row;
+    b_ok[p] = j < K;
+    b_base[p] = b_ok[p] ? (w + (long)j * KRS + (pcol8 ^ prow) * 8) : zpage;
+  }
+
+  auto stage = [&](int buf, int r, int s, int ck) {
+    bf16_t* lA = smem + buf * BUF;
+    bf16_t* lB = lA + BM * BK;
+    const long aoff = ((long)r * W_ + s) * C + ck;
+    #pragma unroll
+    for (int p = 0; p < AP; ++p) {
+      int h = a_hb[p] + r, ww = a_wb[p] + s;
+      bool ok = (a_m[p] < M) & (h >= 0) & (h < H) & (ww >= 0) & (ww < W_);
+      const bf16_t* src = ok ? (a_base[p] + aoff) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src,
+          (LDS_AS void*)(lA + (wave * AP + p) * 512), 16, 0, 0);
+    }
+    const long boff = ((long)r * S + s) * C + ck;
+    #pragma unroll
+    for (int p = 0; p < BP; ++p) {
+      const bf16_t* src = b_ok[p] ? (b_base[p] + boff) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src,
+          (LDS_AS void*)(lB + (wave * BP + p) * 512), 16, 0, 0);
+    }
+  };
+
+  f32x4_v acc[MF][NF];
+  #pragma unroll
+  for (int mi = 0; mi < MF; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < NF; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int fr_row = lane & 15;
+  const int fr_c8 = lane >> 4;
+
+  const int csteps = C / BK;
+  const int T = R * S * csteps;
+  int r2 = 0, s2 = 0, ck2 = 0;
+  auto advance = [&]() {
+    ck2 += BK;
+    if (ck2 >= C) {
+      ck2 = 0;
+      if (++s2 >= S) { s2 = 0; ++r2; }
+    }
+  };
+
+  // prologue: tiles 0 and 1 issued; gate tile 0 landed (tile 1 in flight)
+  stage(0, r2, s2, ck2);
+  advance();
+  if (T > 1) {
+    stage(1, r2, s2, ck2);
+    advance();
+    asm volatile("s_waitcnt vmcnt(%0)" ::"n"(GLDS) : "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < T; ++t) {
+    const bool more = (t + 2) < T;
+    if (more) {
+      stage((t + 2) % 3, r2, s2, ck2);
+      advance();
+    }
+    const bf16_t* lA = smem + (t % 3) * BUF;
+    const bf16_t* lB = lA + BM * BK;
+    #pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      bf16x8_v fa[MF], fb[NF];
+      #pragma unroll
+      for (int mi = 0; mi < MF; ++mi) {
+        int row = wr * WM + mi * 16 + fr_row;
+        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+        fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
+      }
+      #pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        int row = wc * WN + ni * 16 + fr_row;
+        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+        fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
+      }
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int mi = 0; mi < MF; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < NF; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (t + 1 < T) {
+      // tile t+1 must have landed before any wave reads it; tile t+2's
+      // GLDS loads stay in flight across the barrier (never drain to 0)
+      if (more)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(GLDS) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // direct-store epilogue (long K-loop shapes: store tail amortized)
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+  auto out_row = [&](long m) -> long {
+    if (oS == 1) return m;
+    unsigned mu = (unsigned)m;
+    unsigned q1 = mdiv(mu, magic_wo, shift_wo);
+    int wo = (int)(mu - q1 * (unsigned)Wo);
+    unsigned n_u = mdiv(q1, magic_ho, shift_ho);
+    int ho = (int)(q1 - n_u * (unsigned)Ho);
+    return ((long)(int)n_u * oH + (long)ho * oS) * oW + (long)wo * oS;
+  };
+  auto b2f = [](bf16_t u) -> float {
+    union { unsigned i; float f; } cvt;
+    cvt.i = (unsigned)u << 16;
+    return cvt.f;
+  };
+  // accp hoisted OUT of the unrolled loops: a per-element runtime select
+  // around a load makes hipcc branch + wait vmcnt(0) per element
+  auto epilogue = [&](auto has_acc) {
+    #pragma unroll
+    for (int mi = 0; mi < MF; ++mi) {
+      #pragma unroll
+      for (int ni = 0; ni < NF; ++ni) {
+        int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+        if (j >= K) continue;
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          long m = tile_m * BM + wr * WM + mi * 16 + d_row0 + q;
+          if (m < M) {
+            long oi = out_row(m) * K + j;
+            float v = acc[mi][ni][q];
+            if constexpr (decltype(has_acc)::value) v += b2f(accp[oi]);
+            y[oi] = f2b_hw(v);
+          }
+        }
+      }
+    }
+  };
+  if (accp)
+    epilogue(std::true_type{});
+  else
+    epilogue(std::false_type{});
+}
+
+// ---------------------------------------------------------------------------
 // launcher: picks the tile instantiation by shape
 // ---------------------------------------------------------------------------
 static inline long cdiv(long a, long b) { return (a + b - 1) / b; }
@@ -449,6 +683,27 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
   if (mfma_pref < 0) {
     const char* e = getenv("DDLW_CONV_MFMA");
     mfma_pref = (e && e[0] == '3') ? 32 : 16;
+  }
+  // wide-tile pipelined kernel (BM=256, 8 waves, 3-buffer counted-vmcnt
+  // span): on by heuristic for long-K-loop shapes; DDLW_CONV_WIDE=0 off,
+  // =1 force where supported (A/B lever for bench_conv)
+  int wide_pref;  // re-read per call: cheap, and lets bench A/B flip it
+  {
+    const char* e = getenv("DDLW_CONV_WIDE");
+    wide_pref = e ? (e[0] == '0' ? 0 : 1) : -1;  // -1 = auto
+  }
+  {
+    const bool wide_ok = (K >= 128) && (T >= 4);
+    const bool use_wide = wide_ok && (wide_pref == 1 || (wide_pref == -1 && T >= 6));
+    if (use_wide) {
+      long grid = cdiv(M, 256) * cdiv(K, 128);
+      hipLaunchKernelGGL((k_conv_igemm_wide<256, 128>), dim3((int)grid),
+                         dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,
+                         (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,
+                         Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,
+                         sh_wo, mg_ho, sh_ho, (const bf16_t*)acc);
+      DDLW_CHECK_LAUNCH();
+    }
   }
 #define LAUNCH(BM, BN, EPI, BUFS)                                             \
   do {                                                                        \
