@@ -17,7 +17,7 @@
 //     __syncthreads() per step (hipcc adds the vmcnt drain);
 //   - LDS image is lane-linear [rows][64] bf16 (glds writes base+lane*16),
 //     so the bank swizzle lives on the SOURCE address and the fragment
-//     read: chunk col8 ^= (row & 7)  (rule 21: same involution both sides);
+//     read: chunk col8 ^= (row>>1)&7  (rule 21: same involution both sides);
 //   - im2col padding/tile edges are handled by redirecting the per-lane
 //     glds source to a zero page (glds has no predication);
 //   - 2x2 wave grid, each wave (BM/2 x BN/2) of 16x16 fragments,
@@ -120,9 +120,12 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
       int n = (int)n_u;
       a_hb[p] = ho * stride - pad;
       a_wb[p] = wo * stride - pad;
-      // source chunk col8' = dest col8 ^ (row&7) = pcol8 ^ prow (swizzle)
+      // source chunk col8' = dest col8 ^ ((row>>1)&7): the b128 bank row
+      // is 256 B = TWO 128-B LDS rows, so an XOR keyed on row&7 leaves a
+      // 2-way conflict between rows r and r+8 of a 16-row fragment group;
+      // keying on (row>>1)&7 is conflict-free for 16- and 32-row groups
       a_base[p] = x + (((long)n * H + a_hb[p]) * W_ + a_wb[p]) * C +
-                  (pcol8 ^ prow) * 8;
+                  (pcol8 ^ ((((wave * AP + p) & 1) << 2) | (prow >> 1))) * 8;
     } else {
       a_hb[p] = -100000;
       a_wb[p] = -100000;
@@ -137,7 +140,10 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     int row = (wave * BP + p) * 8 + prow;
     int j = tile_n * BN + row;
     b_ok[p] = j < K;
-    b_base[p] = b_ok[p] ? (w + (long)j * KRS + (pcol8 ^ prow) * 8) : zpage;
+    b_base[p] = b_ok[p]
+        ? (w + (long)j * KRS +
+           (pcol8 ^ ((((wave * BP + p) & 1) << 2) | (prow >> 1))) * 8)
+        : zpage;
   }
 
   // ---- staging: one glds per piece into buffer `b` for k-step (r,s,ck)
@@ -221,13 +227,13 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
         #pragma unroll
         for (int mi = 0; mi < MF2; ++mi) {
           int row = wr * WM + mi * 32 + fr_row32;
-          int c8 = (kq * 2 + fr_q8) ^ (row & 7);
+          int c8 = (kq * 2 + fr_q8) ^ ((row >> 1) & 7);
           fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
         }
         #pragma unroll
         for (int ni = 0; ni < NF2; ++ni) {
           int row = wc * WN + ni * 32 + fr_row32;
-          int c8 = (kq * 2 + fr_q8) ^ (row & 7);
+          int c8 = (kq * 2 + fr_q8) ^ ((row >> 1) & 7);
           fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
         }
         #pragma unroll
@@ -244,13 +250,13 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
         #pragma unroll
         for (int mi = 0; mi < MF; ++mi) {
           int row = wr * WM + mi * 16 + fr_row;
-          int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+          int c8 = (kh * 4 + fr_c8) ^ ((row >> 1) & 7);
           fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
         }
         #pragma unroll
         for (int ni = 0; ni < NF; ++ni) {
           int row = wc * WN + ni * 16 + fr_row;
-          int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+          int c8 = (kh * 4 + fr_c8) ^ ((row >> 1) & 7);
           fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
         }
         #pragma unroll
@@ -477,7 +483,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
       a_hb[p] = ho * stride - pad;
       a_wb[p] = wo * stride - pad;
       a_base[p] = x + (((long)n * H + a_hb[p]) * W_ + a_wb[p]) * C +
-                  (pcol8 ^ prow) * 8;
+                  (pcol8 ^ ((((wave * AP + p) & 1) << 2) | (prow >> 1))) * 8;
     } else {
       a_hb[p] = -100000;
       a_wb[p] = -100000;
@@ -492,7 +498,10 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     int row = (wave * BP + p) * 8 + prow;
     int j = tile_n * BN + row;
     b_ok[p] = j < K;
-    b_base[p] = b_ok[p] ? (w + (long)j * KRS + (pcol8 ^ prow) * 8) : zpage;
+    b_base[p] = b_ok[p]
+        ? (w + (long)j * KRS +
+           (pcol8 ^ ((((wave * BP + p) & 1) << 2) | (prow >> 1))) * 8)
+        : zpage;
   }
 
   auto stage = [&](int buf, int r, int s, int ck) {
@@ -564,13 +573,13 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
       #pragma unroll
       for (int mi = 0; mi < MF; ++mi) {
         int row = wr * WM + mi * 16 + fr_row;
-        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+        int c8 = (kh * 4 + fr_c8) ^ ((row >> 1) & 7);
         fa[mi] = *reinterpret_cast<const bf16x8_v*>(lA + row * BK + c8 * 8);
       }
       #pragma unroll
       for (int ni = 0; ni < NF; ++ni) {
         int row = wc * WN + ni * 16 + fr_row;
-        int c8 = (kh * 4 + fr_c8) ^ (row & 7);
+        int c8 = (kh * 4 + fr_c8) ^ ((row >> 1) & 7);
         fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
       }
       __builtin_amdgcn_s_setprio(1);
