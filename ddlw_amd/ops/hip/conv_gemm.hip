@@ -425,7 +425,9 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
 // each piece = 1 glds per wave); the gate before computing tile t+1 is
 // s_waitcnt vmcnt(GLDS) — tile t+2's loads keep flying across the barrier.
 // ---------------------------------------------------------------------------
-template <int BM, int BN>
+template <int BM, int BN, int NB>  // NB = LDS K-tile buffers (3 = span
+                                   // with one tile in flight; 2 = stage
+                                   // hidden behind the current compute)
 __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
     bf16_t* __restrict__ y, const bf16_t* __restrict__ zpage,
@@ -442,8 +444,9 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
   constexpr int AP = BM / 64;         // 1-KiB A pieces per wave (8 rows each)
   constexpr int BP = BN / 64;
   constexpr int GLDS = AP + BP;       // glds per wave per K-tile
+  constexpr int AHEAD = NB - 1;       // tiles staged ahead of the compute
   constexpr int BUF = (BM + BN) * BK; // bf16 elements per K-tile buffer
-  __shared__ __attribute__((aligned(16))) bf16_t smem[3 * BUF];
+  __shared__ __attribute__((aligned(16))) bf16_t smem[NB * BUF];
 
   const long M = (long)N * Ho * Wo;
   const int tiles_n = (K + BN - 1) / BN;
@@ -504,12 +507,17 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
         : zpage;
   }
 
-  auto stage = [&](int buf, int r, int s, int ck) {
+  // staging split into two interleavable halves: glds issue slots are
+  // expensive when burst together (60-185 cyc each in a crowded phase);
+  // spreading them between the two k-half MFMA clusters hides the issue
+  // cost under the matrix pipe (the 8-phase-template idea, coarse form)
+  auto stage_half = [&](int buf, int r, int s, int ck, int half) {
     bf16_t* lA = smem + buf * BUF;
     bf16_t* lB = lA + BM * BK;
     const long aoff = ((long)r * W_ + s) * C + ck;
     #pragma unroll
     for (int p = 0; p < AP; ++p) {
+      if ((p & 1) != half) continue;
       int h = a_hb[p] + r, ww = a_wb[p] + s;
       bool ok = (a_m[p] < M) & (h >= 0) & (h < H) & (ww >= 0) & (ww < W_);
       const bf16_t* src = ok ? (a_base[p] + aoff) : zpage;
@@ -520,6 +528,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     const long boff = ((long)r * S + s) * C + ck;
     #pragma unroll
     for (int p = 0; p < BP; ++p) {
+      if ((p & 1) != half) continue;
       const bf16_t* src = b_ok[p] ? (b_base[p] + boff) : zpage;
       __builtin_amdgcn_global_load_lds(
           (const GLOBAL_AS void*)src,
@@ -547,11 +556,13 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     }
   };
 
-  // prologue: tiles 0 and 1 issued; gate tile 0 landed (tile 1 in flight)
-  stage(0, r2, s2, ck2);
+  // prologue: tiles 0..AHEAD-1 issued; gate tile 0 landed
+  stage_half(0, r2, s2, ck2, 0);
+  stage_half(0, r2, s2, ck2, 1);
   advance();
-  if (T > 1) {
-    stage(1, r2, s2, ck2);
+  if (AHEAD > 1 && T > 1) {
+    stage_half(1, r2, s2, ck2, 0);
+    stage_half(1, r2, s2, ck2, 1);
     advance();
     asm volatile("s_waitcnt vmcnt(%0)" ::"n"(GLDS) : "memory");
   } else {
@@ -560,15 +571,14 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < T; ++t) {
-    const bool more = (t + 2) < T;
-    if (more) {
-      stage((t + 2) % 3, r2, s2, ck2);
-      advance();
-    }
-    const bf16_t* lA = smem + (t % 3) * BUF;
+    const bool more = (t + AHEAD) < T;
+    const int nbuf = (t + AHEAD) % NB, nr = r2, ns = s2, nck = ck2;
+    if (more) advance();
+    const bf16_t* lA = smem + (t % NB) * BUF;
     const bf16_t* lB = lA + BM * BK;
     #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
+      if (more) stage_half(nbuf, nr, ns, nck, kh);
       bf16x8_v fa[MF], fb[NF];
       #pragma unroll
       for (int mi = 0; mi < MF; ++mi) {
@@ -592,9 +602,9 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
       __builtin_amdgcn_s_setprio(0);
     }
     if (t + 1 < T) {
-      // tile t+1 must have landed before any wave reads it; tile t+2's
-      // GLDS loads stay in flight across the barrier (never drain to 0)
-      if (more)
+      // tile t+1 must have landed before any wave reads it; with NB=3 the
+      // newest tile's GLDS loads stay in flight across the barrier
+      if (AHEAD > 1 && more)
         asm volatile("s_waitcnt vmcnt(%0)" ::"n"(GLDS) : "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -702,15 +712,32 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
     wide_pref = e ? (e[0] == '0' ? 0 : 1) : -1;  // -1 = auto
   }
   {
-    const bool wide_ok = (K >= 128) && (T >= 4);
-    const bool use_wide = wide_ok && (wide_pref == 1 || (wide_pref == -1 && T >= 6));
+    // measured routing (bench/tools/wide_check.py on MI355X): the 256x256
+    // 2-buf wide kernel wins (+20-28%) where K >= 256, the K-loop is deep
+    // enough to amortize its prologue (T >= 9), and the grid still fills
+    // most CUs (>= 150 blocks); everywhere else the 128x128 2-block/CU
+    // kernel wins on occupancy. The 256x128 3-buf variant never won.
+    const long wide_blocks = cdiv(M, 256) * cdiv(K, 256);
+    const bool wide_auto = (K >= 256) && (T >= 9) && (wide_blocks >= 150);
+    const bool use_wide =
+        (wide_pref == 1 && K >= 128 && T >= 4) ||
+        (wide_pref == -1 && wide_auto);
     if (use_wide) {
-      long grid = cdiv(M, 256) * cdiv(K, 128);
-      hipLaunchKernelGGL((k_conv_igemm_wide<256, 128>), dim3((int)grid),
-                         dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,
-                         (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,
-                         Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,
-                         sh_wo, mg_ho, sh_ho, (const bf16_t*)acc);
+      if (K >= 256) {
+        long grid = cdiv(M, 256) * cdiv(K, 256);
+        hipLaunchKernelGGL((k_conv_igemm_wide<256, 256, 2>), dim3((int)grid),
+                           dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,
+                           (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,
+                           Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,
+                           sh_wo, mg_ho, sh_ho, (const bf16_t*)acc);
+      } else {
+        long grid = cdiv(M, 256) * cdiv(K, 128);
+        hipLaunchKernelGGL((k_conv_igemm_wide<256, 128, 3>), dim3((int)grid),
+                           dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,
+                           (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,
+                           Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,
+                           sh_wo, mg_ho, sh_ho, (const bf16_t*)acc);
+      }
       DDLW_CHECK_LAUNCH();
     }
   }
