@@ -70,6 +70,25 @@ def bn_stats(x: torch.Tensor, eps: float, momentum: float,
     return mean, rstd
 
 
+def bn_finalize_parts(part_sum: torch.Tensor, part_sumsq: torch.Tensor,
+                      nparts: int, rows: int, C: int, eps: float,
+                      momentum: float, running_mean, running_var):
+    """Finalize mean/rstd from conv-epilogue-fused partial sums (the
+    k_bn_stats read pass is skipped entirely)."""
+    lib = require_lib()
+    dev = part_sum.device
+    mean = torch.empty(C, dtype=torch.float32, device=dev)
+    rstd = torch.empty(C, dtype=torch.float32, device=dev)
+    check(lib.ddlw_bn_finalize_n(_p(part_sum), _p(part_sumsq), _p(mean),
+                                 _p(rstd), _p(running_mean), _p(running_var),
+                                 ctypes.c_long(rows), ctypes.c_int(C),
+                                 ctypes.c_int(nparts), ctypes.c_float(eps),
+                                 ctypes.c_float(momentum),
+                                 ctypes.c_void_p(current_stream_ptr())),
+          "bn_finalize_n")
+    return mean, rstd
+
+
 def bn_apply(x: torch.Tensor, res: Optional[torch.Tensor], mean, rstd, gamma, beta,
              relu: bool):
     """Returns (y, mask): mask is a uint8 tensor of rows*C/8 relu bits (one

@@ -68,6 +68,25 @@ def _fwd_conv(x, w, stride, padding, hip_fwd):
     return torch.nn.functional.conv2d(x, w.to(x.dtype), None, stride, padding)
 
 
+def _fwd_conv_bn_stats(x, w, stride, padding, hip_fwd, bn):
+    """conv + BN batch statistics. On the hip path the conv epilogue emits
+    per-tile partial sums (no separate k_bn_stats read of the conv output);
+    otherwise conv then the standalone stats pass. Returns (t, mean, rstd).
+    DDLW_FUSED_BN_STATS=0 disables the fusion (A/B lever)."""
+    if hip_fwd and os.environ.get("DDLW_FUSED_BN_STATS", "1") == "1":
+        t, parts, np_ = conv_gemm.conv_fwd_kernel(
+            x, w.to(torch.bfloat16), stride, padding, bn_parts=True)
+        rows = t.shape[0] * t.shape[2] * t.shape[3]
+        mean, rstd = binding.bn_finalize_parts(
+            parts[0], parts[1], np_, rows, t.shape[1], bn.eps, bn.momentum,
+            bn.running_mean, bn.running_var)
+        return t, mean, rstd
+    t = _fwd_conv(x, w, stride, padding, hip_fwd)
+    mean, rstd = binding.bn_stats(t, bn.eps, bn.momentum,
+                                  bn.running_mean, bn.running_var)
+    return t, mean, rstd
+
+
 class _BottleneckFn(torch.autograd.Function):
     """Training-mode fused bottleneck. ``wd/gd/bd`` are the downsample
     conv/BN parameters (None for identity-shortcut blocks)."""
@@ -84,9 +103,7 @@ class _BottleneckFn(torch.autograd.Function):
             dsbn = block.downsample.bn
             conv_gemm.available(dsc, x, mode)
             rdf, rdd, rdg = getattr(dsc, "_ddlw_route", (False, False, False))
-            td = _fwd_conv(x, wd, stride, 0, rdf)
-            md, sd = binding.bn_stats(td, dsbn.eps, dsbn.momentum,
-                                      dsbn.running_mean, dsbn.running_var)
+            td, md, sd = _fwd_conv_bn_stats(x, wd, stride, 0, rdf, dsbn)
             res, _ = binding.bn_apply(td, None, md, sd, gd, bd, False)
             dsbn.num_batches_tracked += 1
         else:
@@ -96,23 +113,17 @@ class _BottleneckFn(torch.autograd.Function):
 
         conv_gemm.available(block.conv1, x, mode)
         r1f, r1d, r1g = getattr(block.conv1, "_ddlw_route", (False, False, False))
-        t1 = _fwd_conv(x, w1, 1, 0, r1f)
-        m1, s1 = binding.bn_stats(t1, bn1.eps, bn1.momentum,
-                                  bn1.running_mean, bn1.running_var)
+        t1, m1, s1 = _fwd_conv_bn_stats(x, w1, 1, 0, r1f, bn1)
         a1, mask1 = binding.bn_apply(t1, None, m1, s1, g1, b1, True)
 
         conv_gemm.available(block.conv2, a1, mode)
         r2f, r2d, r2g = getattr(block.conv2, "_ddlw_route", (False, False, False))
-        t2 = _fwd_conv(a1, w2, stride, 1, r2f)
-        m2, s2 = binding.bn_stats(t2, bn2.eps, bn2.momentum,
-                                  bn2.running_mean, bn2.running_var)
+        t2, m2, s2 = _fwd_conv_bn_stats(a1, w2, stride, 1, r2f, bn2)
         a2, mask2 = binding.bn_apply(t2, None, m2, s2, g2, b2, True)
 
         conv_gemm.available(block.conv3, a2, mode)
         r3f, r3d, r3g = getattr(block.conv3, "_ddlw_route", (False, False, False))
-        t3 = _fwd_conv(a2, w3, 1, 0, r3f)
-        m3, s3 = binding.bn_stats(t3, bn3.eps, bn3.momentum,
-                                  bn3.running_mean, bn3.running_var)
+        t3, m3, s3 = _fwd_conv_bn_stats(a2, w3, 1, 0, r3f, bn3)
         out, mask3 = binding.bn_apply(t3, res, m3, s3, g3, b3, True)
 
         for bn in (bn1, bn2, bn3):

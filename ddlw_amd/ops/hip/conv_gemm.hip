@@ -51,6 +51,57 @@ __device__ __forceinline__ bf16_t f2b_hw(float f) {
   return cvt.u;
 }
 
+// BN-stats fusion: the conv epilogue already holds every output value in
+// registers — accumulating per-channel sum/sumsq here saves the separate
+// k_bn_stats full read of the conv output (~1.2 ms/step on ResNet-50 b256).
+// Each (tile_m, wr) wave-row writes ONE deterministic partial row
+// [nparts][K]; nparts = grid_m * (BM/WM); k_bn_finalize reduces them.
+// Values are accumulated bf16-rounded so the stats match the unfused
+// k_bn_stats(y) numerics exactly.
+template <int MF, int NF, int BM, int BN, int WM, int WN, typename ACC>
+__device__ __forceinline__ void bn_partials_16(
+    const ACC& acc, float* __restrict__ bn_ps, float* __restrict__ bn_pq,
+    long tile_m, int tile_n, int wr, int wc, long M, int K) {
+  const int lane = threadIdx.x & 63;
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+  float s[NF], q[NF];
+  #pragma unroll
+  for (int ni = 0; ni < NF; ++ni) { s[ni] = 0.f; q[ni] = 0.f; }
+  #pragma unroll
+  for (int mi = 0; mi < MF; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < NF; ++ni)
+      #pragma unroll
+      for (int qq = 0; qq < 4; ++qq) {
+        long m = tile_m * BM + (long)wr * WM + mi * 16 + d_row0 + qq;
+        if (m < M) {
+          union { unsigned i; float f; } cv;
+          cv.i = (unsigned)f2b_hw(acc[mi][ni][qq]) << 16;
+          s[ni] += cv.f;
+          q[ni] += cv.f * cv.f;
+        }
+      }
+  #pragma unroll
+  for (int ni = 0; ni < NF; ++ni) {
+    s[ni] += __shfl_xor(s[ni], 16, 64);
+    s[ni] += __shfl_xor(s[ni], 32, 64);
+    q[ni] += __shfl_xor(q[ni], 16, 64);
+    q[ni] += __shfl_xor(q[ni], 32, 64);
+  }
+  if (lane < 16) {
+    const long prow_ = tile_m * (BM / WM) + wr;
+    #pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
+      int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+      if (j < K) {
+        bn_ps[prow_ * K + j] = s[ni];
+        bn_pq[prow_ * K + j] = q[ni];
+      }
+    }
+  }
+}
+
 template <int BM, int BN, bool EPI_LDS, int BUFS, bool M32EN = true>
 __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
@@ -60,7 +111,9 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     int oH, int oW, int oS,  // output scatter: flat out row = (n*oH + ho*oS)*oW + wo*oS
     unsigned long long magic_wo, unsigned shift_wo,
     unsigned long long magic_ho, unsigned shift_ho,
-    const bf16_t* __restrict__ accp) {  // optional epilogue accumulate input
+    const bf16_t* __restrict__ accp,   // optional epilogue accumulate input
+    float* __restrict__ bn_ps,         // optional BN partial sums [nparts][K]
+    float* __restrict__ bn_pq) {       // optional BN partial sumsq
   // accp != nullptr: y = conv + accp (read at the output index). Used to
   // fuse the residual-join gradient add (d_block_input = conv1_dgrad +
   // d_identity) into the dgrad epilogue — saves the engine's separate
@@ -337,6 +390,7 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           }
         }
       }
+      if (bn_ps) bn_partials_16<MF, NF, BM, BN, WM, WN>(acc, bn_ps, bn_pq, tile_m, tile_n, wr, wc, M, K);
     }
     return;
   }
@@ -436,7 +490,8 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     int oH, int oW, int oS,
     unsigned long long magic_wo, unsigned shift_wo,
     unsigned long long magic_ho, unsigned shift_ho,
-    const bf16_t* __restrict__ accp) {
+    const bf16_t* __restrict__ accp,
+    float* __restrict__ bn_ps, float* __restrict__ bn_pq) {
   constexpr int BK = 64;
   constexpr int WAVES = 8;            // 4 (M) x 2 (N)
   constexpr int WM = BM / 4, WN = BN / 2;
@@ -656,6 +711,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     epilogue(std::true_type{});
   else
     epilogue(std::false_type{});
+  if (bn_ps) bn_partials_16<MF, NF, BM, BN, WM, WN>(acc, bn_ps, bn_pq, tile_m, tile_n, wr, wc, M, K);
 }
 
 // ---------------------------------------------------------------------------
@@ -673,12 +729,62 @@ static inline void make_magic(unsigned d, unsigned long long* magic, unsigned* s
   *shift = 32 + s;
 }
 
-DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
-                                        const void* zpage, const void* acc,
-                                        int N, int H, int W_, int C, int K,
-                                        int Ho, int Wo, int R, int S, int stride,
-                                        int pad, int oH, int oW, int oS,
-                                        void* stream) {
+struct TilePick {
+  int bm, bn, bufs, wvm;
+  bool epi_lds, wide, m32;
+};
+
+// One tile decision for launcher + nparts query (must stay in sync by
+// construction). want_stats forces the 16x16 direct epilogue (the only
+// ones carrying the bn-partials code).
+static TilePick pick_tile(long M, int K, int C, int T, bool want_stats) {
+  int mfma_pref, wide_pref;
+  {
+    const char* e = getenv("DDLW_CONV_MFMA");
+    mfma_pref = (e && e[0] == '3') ? 32 : 16;
+    const char* e2 = getenv("DDLW_CONV_WIDE");
+    wide_pref = e2 ? (e2[0] == '0' ? 0 : 1) : -1;  // -1 = auto
+  }
+  TilePick p{};
+  p.m32 = (mfma_pref == 32) && !want_stats;
+  // measured routing (bench/tools/wide_check.py on MI355X): the 256x256
+  // 2-buf wide kernel wins (+20-28%) where K >= 256, the K-loop is deep
+  // enough to amortize its prologue (T >= 9), and the grid still fills
+  // most CUs (>= 150 blocks); everywhere else the 128x128 2-block/CU
+  // kernel wins on occupancy. The 256x128 3-buf variant never won an
+  // auto route (kept for DDLW_CONV_WIDE=1 A/Bs).
+  const long wide_blocks = cdiv(M, 256) * cdiv(K, 256);
+  const bool wide_auto = (K >= 256) && (T >= 9) && (wide_blocks >= 150);
+  if ((wide_pref == 1 && K >= 128 && T >= 4) || (wide_pref == -1 && wide_auto)) {
+    p.wide = true;
+    p.bm = 256;
+    p.bn = (K >= 256) ? 256 : 128;
+    p.bufs = (p.bn == 256) ? 2 : 3;
+    p.wvm = 4;
+    return p;
+  }
+  p.bm = 128;
+  p.wvm = 2;
+  p.bn = (K >= 128) ? 128 : (K >= 64 ? 64 : 32);
+  p.epi_lds = (T <= 4) && !want_stats;
+  p.bufs = (T == 1) ? 1 : 2;
+  return p;
+}
+
+DDLW_EXPORT long ddlw_conv_fwd_nparts(int N, int C, int K, int Ho, int Wo,
+                                      int R, int S) {
+  long M = (long)N * Ho * Wo;
+  const int T = R * S * (C / 64);
+  TilePick p = pick_tile(M, K, C, T, true);
+  return cdiv(M, p.bm) * p.wvm;
+}
+
+static int conv_fwd_launch(const void* x, const void* w, void* y,
+                           const void* zpage, const void* acc,
+                           void* bn_ps, void* bn_pq,
+                           int N, int H, int W_, int C, int K,
+                           int Ho, int Wo, int R, int S, int stride,
+                           int pad, int oH, int oW, int oS, void* stream) {
   if (C % 64 != 0) {
     ddlw_set_error("conv_fwd_igemm: C must be a multiple of 64");
     return 2;
@@ -694,86 +800,89 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
   make_magic((unsigned)Ho, &mg_ho, &sh_ho);
   const int T = R * S * (C / 64);
   hipStream_t st = (hipStream_t)stream;
-  // measured within-box A/B: the 16x16x32 path is 6-9% faster than
-  // 32x32x16 on the ResNet shapes (16 independent accumulator chains vs 4 —
-  // the 32x32 dependent-accumulator latency is not hidden at 2 waves/SIMD);
-  // DDLW_CONV_MFMA=32 keeps the alternative for future A/Bs.
-  static int mfma_pref = -1;
-  if (mfma_pref < 0) {
-    const char* e = getenv("DDLW_CONV_MFMA");
-    mfma_pref = (e && e[0] == '3') ? 32 : 16;
-  }
-  // wide-tile pipelined kernel (BM=256, 8 waves, 3-buffer counted-vmcnt
-  // span): on by heuristic for long-K-loop shapes; DDLW_CONV_WIDE=0 off,
-  // =1 force where supported (A/B lever for bench_conv)
-  int wide_pref;  // re-read per call: cheap, and lets bench A/B flip it
-  {
-    const char* e = getenv("DDLW_CONV_WIDE");
-    wide_pref = e ? (e[0] == '0' ? 0 : 1) : -1;  // -1 = auto
-  }
-  {
-    // measured routing (bench/tools/wide_check.py on MI355X): the 256x256
-    // 2-buf wide kernel wins (+20-28%) where K >= 256, the K-loop is deep
-    // enough to amortize its prologue (T >= 9), and the grid still fills
-    // most CUs (>= 150 blocks); everywhere else the 128x128 2-block/CU
-    // kernel wins on occupancy. The 256x128 3-buf variant never won.
-    const long wide_blocks = cdiv(M, 256) * cdiv(K, 256);
-    const bool wide_auto = (K >= 256) && (T >= 9) && (wide_blocks >= 150);
-    const bool use_wide =
-        (wide_pref == 1 && K >= 128 && T >= 4) ||
-        (wide_pref == -1 && wide_auto);
-    if (use_wide) {
-      if (K >= 256) {
-        long grid = cdiv(M, 256) * cdiv(K, 256);
-        hipLaunchKernelGGL((k_conv_igemm_wide<256, 256, 2>), dim3((int)grid),
-                           dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,
-                           (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,
-                           Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,
-                           sh_wo, mg_ho, sh_ho, (const bf16_t*)acc);
-      } else {
-        long grid = cdiv(M, 256) * cdiv(K, 128);
-        hipLaunchKernelGGL((k_conv_igemm_wide<256, 128, 3>), dim3((int)grid),
-                           dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,
-                           (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,
-                           Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,
-                           sh_wo, mg_ho, sh_ho, (const bf16_t*)acc);
-      }
-      DDLW_CHECK_LAUNCH();
-    }
+  const bool want_stats = bn_ps != nullptr;
+  TilePick p = pick_tile(M, K, C, T, want_stats);
+  if (p.wide) {
+    long grid = cdiv(M, 256) * cdiv(K, p.bn);
+#define WIDE_LAUNCH(BN_, NB_)                                                  \
+    hipLaunchKernelGGL((k_conv_igemm_wide<256, BN_, NB_>), dim3((int)grid),    \
+                       dim3(512), 0, st, (const bf16_t*)x, (const bf16_t*)w,   \
+                       (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,   \
+                       Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,    \
+                       sh_wo, mg_ho, sh_ho, (const bf16_t*)acc,                \
+                       (float*)bn_ps, (float*)bn_pq)
+    if (p.bn == 256)
+      WIDE_LAUNCH(256, 2);
+    else
+      WIDE_LAUNCH(128, 3);
+#undef WIDE_LAUNCH
+    DDLW_CHECK_LAUNCH();
   }
 #define LAUNCH(BM, BN, EPI, BUFS)                                             \
   do {                                                                        \
     long grid = cdiv(M, BM) * cdiv(K, BN);                                    \
-    if (mfma_pref == 16)                                                      \
+    if (!p.m32)                                                               \
       hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, false>),        \
                          dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
                          (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
                          N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
                          (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho,   \
-                         (const bf16_t*)acc);                                 \
+                         (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq);   \
     else                                                                      \
       hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, true>),         \
                          dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
                          (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
                          N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
                          (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho,   \
-                         (const bf16_t*)acc);                                 \
+                         (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq);   \
   } while (0)
-  if (K >= 128) {
-    if (T == 1) LAUNCH(128, 128, true, 1);
-    else if (T <= 4) LAUNCH(128, 128, true, 2);
-    else LAUNCH(128, 128, false, 2);
-  } else if (K >= 64) {
-    if (T == 1) LAUNCH(128, 64, true, 1);
-    else if (T <= 4) LAUNCH(128, 64, true, 2);
-    else LAUNCH(128, 64, false, 2);
+  if (p.bn == 128) {
+    if (p.epi_lds) {
+      if (p.bufs == 1) LAUNCH(128, 128, true, 1);
+      else LAUNCH(128, 128, true, 2);
+    } else {
+      if (p.bufs == 1) LAUNCH(128, 128, false, 1);
+      else LAUNCH(128, 128, false, 2);
+    }
+  } else if (p.bn == 64) {
+    if (p.epi_lds) {
+      if (p.bufs == 1) LAUNCH(128, 64, true, 1);
+      else LAUNCH(128, 64, true, 2);
+    } else {
+      if (p.bufs == 1) LAUNCH(128, 64, false, 1);
+      else LAUNCH(128, 64, false, 2);
+    }
   } else {
-    if (T == 1) LAUNCH(128, 32, true, 1);
-    else if (T <= 4) LAUNCH(128, 32, true, 2);
-    else LAUNCH(128, 32, false, 2);
+    if (p.epi_lds) {
+      if (p.bufs == 1) LAUNCH(128, 32, true, 1);
+      else LAUNCH(128, 32, true, 2);
+    } else {
+      if (p.bufs == 1) LAUNCH(128, 32, false, 1);
+      else LAUNCH(128, 32, false, 2);
+    }
   }
 #undef LAUNCH
   DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
+                                        const void* zpage, const void* acc,
+                                        int N, int H, int W_, int C, int K,
+                                        int Ho, int Wo, int R, int S, int stride,
+                                        int pad, int oH, int oW, int oS,
+                                        void* stream) {
+  return conv_fwd_launch(x, w, y, zpage, acc, nullptr, nullptr, N, H, W_, C,
+                         K, Ho, Wo, R, S, stride, pad, oH, oW, oS, stream);
+}
+
+DDLW_EXPORT int ddlw_conv_fwd_igemm_stats(const void* x, const void* w, void* y,
+                                          const void* zpage,
+                                          void* bn_ps, void* bn_pq,
+                                          int N, int H, int W_, int C, int K,
+                                          int Ho, int Wo, int R, int S,
+                                          int stride, int pad, void* stream) {
+  return conv_fwd_launch(x, w, y, zpage, nullptr, bn_ps, bn_pq, N, H, W_, C,
+                         K, Ho, Wo, R, S, stride, pad, Ho, Wo, 1, stream);
 }
 
 DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,

@@ -703,6 +703,20 @@ DDLW_EXPORT int ddlw_bn_finalize(const void* part_sum, const void* part_sumsq,
   DDLW_CHECK_LAUNCH();
 }
 
+// finalize over an EXPLICIT partial count (conv-epilogue fused stats write
+// grid_m * waves_m partial rows, not ddlw_bn_nparts's geometry)
+DDLW_EXPORT int ddlw_bn_finalize_n(const void* part_sum, const void* part_sumsq,
+                                   void* mean, void* rstd, void* rmean,
+                                   void* rvar, long rows, int C, int nparts,
+                                   float eps, float momentum, void* stream) {
+  hipLaunchKernelGGL(k_bn_finalize, dim3((C + 31) / 32), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)part_sum,
+                     (const float*)part_sumsq, (float*)mean, (float*)rstd,
+                     (float*)rmean, (float*)rvar, rows, C, nparts, eps,
+                     momentum);
+  DDLW_CHECK_LAUNCH();
+}
+
 
 // grid for the fixed-channel elementwise BN kernels: x covers channel-vector
 // groups, y covers row groups (capped; kernels stride the remainder)

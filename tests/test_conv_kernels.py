@@ -269,3 +269,40 @@ def test_fused_bottleneck_vs_fp32_oracle(monkeypatch):
     assert ((out.float().cpu() - out_o).abs().max() / s).item() < 6e-2
     so = xo.grad.abs().max().item() + 1e-6
     assert ((xa.grad.float().cpu() - xo.grad).abs().max() / so).item() < 1.5e-1
+
+
+@pytest.mark.parametrize("shape", [
+    (14, 14, 64, 128, 1, 1, 1),     # T=1 (EPI-direct forced by stats)
+    (14, 14, 128, 64, 3, 3, 1),     # deep K-loop
+    (28, 28, 256, 256, 3, 3, 1),    # wide-kernel route (K>=256, T>=9)
+    (14, 14, 192, 128, 1, 1, 1),    # K not multiple of BN tile? (K=128 fine) C=192 T=3
+])
+def test_conv_fused_bn_stats_parity(shape):
+    """The conv-epilogue fused BN partials must reproduce the standalone
+    k_bn_stats(y) mean/rstd (same bf16-rounded accumulation)."""
+    from ddlw_amd.ops import binding, conv_gemm
+
+    H, W, C, K, R, S, st = shape
+    pad = 1 if R == 3 else 0
+    torch.manual_seed(2)
+    B = 37  # odd batch: partial last tile exercises the m < M guard
+    x = _cl(torch.randn(B, C, H, W, device=_cuda()).to(torch.bfloat16))
+    w = _cl(torch.randn(K, C, R, S, device=_cuda()).to(torch.bfloat16) * 0.1)
+    y, parts, np_ = conv_gemm.conv_fwd_kernel(x, w, st, pad, bn_parts=True)
+    rows = y.shape[0] * y.shape[2] * y.shape[3]
+    rm = torch.zeros(K, device=_cuda())
+    rv = torch.ones(K, device=_cuda())
+    mean, rstd = binding.bn_finalize_parts(
+        parts[0], parts[1], np_, rows, K, 1e-5, 0.1, rm, rv)
+    rm2 = torch.zeros(K, device=_cuda())
+    rv2 = torch.ones(K, device=_cuda())
+    mean2, rstd2 = binding.bn_stats(y, 1e-5, 0.1, rm2, rv2)
+    assert torch.allclose(mean, mean2, atol=1e-3, rtol=1e-3), \
+        (mean - mean2).abs().max()
+    assert torch.allclose(rstd, rstd2, atol=1e-3, rtol=1e-3), \
+        (rstd - rstd2).abs().max()
+    assert torch.allclose(rm, rm2, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(rv, rv2, atol=1e-3, rtol=1e-3)
+    # the conv output itself unchanged vs the stats-free launch
+    y2 = conv_gemm.conv_fwd_kernel(x, w, st, pad)
+    assert torch.equal(y, y2)
