@@ -424,6 +424,9 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   const int e_ch = lane % CPL;       // chunk within the row
   const long m_base = tile_m * BM + wr * WM;
   const int j_base = tile_n * BN + wc * WN + e_ch * 8;
+  float s8[8], q8[8];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) { s8[e] = 0.f; q8[e] = 0.f; }
   #pragma unroll
   for (int it = 0; it < WM / RPI; ++it) {
     const int row = it * RPI + e_row;
@@ -448,6 +451,16 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           val.w = addpair(val.w, a.w);
         }
         *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
+        if (bn_ps) {
+          // fused BN partials (pure VALU; never combined with accp)
+          #pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
+            float f = b2f((bf16_t)(wd >> ((e & 1) * 16)));
+            s8[e] += f;
+            q8[e] += f * f;
+          }
+        }
       } else {
         // static component extraction (a reinterpret pointer into `val`
         // forces the register to scratch — rule 20)
@@ -459,7 +472,34 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
             if (accp)
               ov = f2b_hw(b2f(ov) + b2f(accp[orow * K + j_base + e]));
             y[orow * K + j_base + e] = ov;
+            if (bn_ps) {
+              float f = b2f(ov);
+              s8[e] += f;
+              q8[e] += f * f;
+            }
           }
+        }
+      }
+    }
+  }
+  if (bn_ps) {
+    // lanes sharing e_ch (= lane % CPL) hold disjoint row groups of the
+    // same 8 channels: xor-reduce over the row bits, lanes < CPL write
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      for (int msk = CPL; msk < 64; msk <<= 1) {
+        s8[e] += __shfl_xor(s8[e], msk, 64);
+        q8[e] += __shfl_xor(q8[e], msk, 64);
+      }
+    }
+    if (lane < CPL) {
+      const long prow_ = tile_m * 2 + wr;  // WVM = BM/WM = 2
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int j = j_base + e;
+        if (j < K) {
+          bn_ps[prow_ * K + j] = s8[e];
+          bn_pq[prow_ * K + j] = q8[e];
         }
       }
     }
@@ -766,7 +806,7 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats) {
   p.bm = 128;
   p.wvm = 2;
   p.bn = (K >= 128) ? 128 : (K >= 64 ? 64 : 32);
-  p.epi_lds = (T <= 4) && !want_stats;
+  p.epi_lds = (T <= 4);
   p.bufs = (T == 1) ? 1 : 2;
   return p;
 }
