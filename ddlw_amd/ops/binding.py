@@ -79,6 +79,18 @@ def bn_finalize_parts(part_sum: torch.Tensor, part_sumsq: torch.Tensor,
     dev = part_sum.device
     mean = torch.empty(C, dtype=torch.float32, device=dev)
     rstd = torch.empty(C, dtype=torch.float32, device=dev)
+    if nparts > 512:
+        # chip-filling fold first: k_bn_finalize's grid is ceil(C/32)
+        # blocks and would serialize a 12k-row partial read on 2 CUs
+        G = max(8, 512 // max(1, (C + 31) // 32))
+        stage = torch.empty(2, G, C, dtype=torch.float32, device=dev)
+        check(lib.ddlw_bn_parts_fold(_p(part_sum), _p(part_sumsq),
+                                     _p(stage[0]), _p(stage[1]),
+                                     ctypes.c_long(nparts), ctypes.c_int(C),
+                                     ctypes.c_int(G),
+                                     ctypes.c_void_p(current_stream_ptr())),
+              "bn_parts_fold")
+        part_sum, part_sumsq, nparts = stage[0], stage[1], G
     check(lib.ddlw_bn_finalize_n(_p(part_sum), _p(part_sumsq), _p(mean),
                                  _p(rstd), _p(running_mean), _p(running_var),
                                  ctypes.c_long(rows), ctypes.c_int(C),

@@ -703,6 +703,50 @@ DDLW_EXPORT int ddlw_bn_finalize(const void* part_sum, const void* part_sumsq,
   DDLW_CHECK_LAUNCH();
 }
 
+// Fold a LARGE partial set (conv-epilogue fused stats: nparts = grid_m x
+// waves_m, up to ~12k rows) down to G rows with a chip-filling grid —
+// k_bn_finalize's grid is only ceil(C/32) blocks (2 for C=64) and would
+// serialize the strided reads of a 12k-row partial buffer on 2 CUs.
+__global__ __launch_bounds__(256) void k_bn_parts_fold(
+    const float* __restrict__ ps, const float* __restrict__ pq,
+    float* __restrict__ os, float* __restrict__ oq, long nparts, int C,
+    int G) {
+  const int cl = threadIdx.x & 31;
+  const int pg = threadIdx.x >> 5;
+  const int c = blockIdx.x * 32 + cl;
+  const int g = blockIdx.y;
+  float s = 0.f, q = 0.f;
+  if (c < C) {
+    for (long p = (long)g * 8 + pg; p < nparts; p += (long)G * 8) {
+      s += ps[p * C + c];
+      q += pq[p * C + c];
+    }
+  }
+  __shared__ float ls[8][32], lq[8][32];
+  ls[pg][cl] = s;
+  lq[pg][cl] = q;
+  __syncthreads();
+  if (pg == 0 && c < C) {
+    #pragma unroll
+    for (int i = 1; i < 8; ++i) {
+      s += ls[i][cl];
+      q += lq[i][cl];
+    }
+    os[(long)g * C + c] = s;
+    oq[(long)g * C + c] = q;
+  }
+}
+
+DDLW_EXPORT int ddlw_bn_parts_fold(const void* ps, const void* pq, void* os,
+                                   void* oq, long nparts, int C, int G,
+                                   void* stream) {
+  dim3 grid((C + 31) / 32, G);
+  hipLaunchKernelGGL(k_bn_parts_fold, grid, dim3(256), 0, (hipStream_t)stream,
+                     (const float*)ps, (const float*)pq, (float*)os,
+                     (float*)oq, nparts, C, G);
+  DDLW_CHECK_LAUNCH();
+}
+
 // finalize over an EXPLICIT partial count (conv-epilogue fused stats write
 // grid_m * waves_m partial rows, not ddlw_bn_nparts's geometry)
 DDLW_EXPORT int ddlw_bn_finalize_n(const void* part_sum, const void* part_sumsq,
