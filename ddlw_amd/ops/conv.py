@@ -49,6 +49,21 @@ class Conv2d(nn.Conv2d):
             return binding.depthwise_fwd(
                 x.contiguous(memory_format=torch.channels_last), self.weight, st, pd
             )
+        if (
+            hip_ok
+            and self.groups == 1
+            and self.in_channels == 3
+            and self.out_channels == 64
+            and self.kernel_size == (7, 7)
+            and self.stride == (2, 2)
+            and self.padding == (3, 3)
+            and self.bias is None
+            and os.environ.get("DDLW_STEM", "1") == "1"
+        ):
+            # dedicated C=3 stem kernel (K1): c4+halo repack + 2x8x4 igemm
+            from . import conv_gemm
+
+            return conv_gemm.stem_conv2d(x, self.weight)
         if hip_ok and self.groups == 1:
             from . import conv_gemm
 

@@ -311,3 +311,32 @@ def test_conv_fused_bn_stats_parity(shape):
     # the conv output itself unchanged vs the stats-free launch
     y2 = conv_gemm.conv_fwd_kernel(x, w, st, pad)
     assert torch.equal(y, y2)
+
+
+def test_stem_conv_parity():
+    """Dedicated C=3 stem kernel (7x7 s2 p3) vs the fp32 stock op."""
+    from ddlw_amd.ops import conv_gemm
+
+    torch.manual_seed(3)
+    x = _cl(torch.randn(5, 3, 224, 224, device=_cuda()).to(torch.bfloat16))
+    w = _cl(torch.randn(64, 3, 7, 7, device=_cuda()).to(torch.bfloat16) * 0.2)
+    y = conv_gemm.stem_fwd_kernel(x, w).float()
+    ref = F.conv2d(x.float(), w.float(), None, 2, 3)
+    assert y.shape == ref.shape
+    scale = ref.abs().max() + 1e-6
+    assert ((y - ref).abs().max() / scale).item() < 5e-2
+
+
+def test_stem_conv_autograd_route():
+    """Conv2d module routes the stem shape to the dedicated kernel and the
+    backward (library dgrad/wgrad) still flows."""
+    from ddlw_amd.ops.conv import Conv2d
+
+    conv = Conv2d(3, 64, 7, stride=2, padding=3, bias=False).to(_cuda())
+    conv = conv.to(memory_format=torch.channels_last).to(torch.bfloat16)
+    x = _cl(torch.randn(2, 3, 64, 64, device=_cuda()).to(torch.bfloat16))
+    y = conv(x)
+    assert y.grad_fn is not None and "Stem" in type(y.grad_fn).__name__
+    y.float().square().mean().backward()
+    assert conv.weight.grad is not None
+    assert torch.isfinite(conv.weight.grad.float()).all()
