@@ -101,6 +101,30 @@ def bn_finalize_parts(part_sum: torch.Tensor, part_sumsq: torch.Tensor,
     return mean, rstd
 
 
+def bn_grad_finalize_parts(part_db: torch.Tensor, part_dg: torch.Tensor,
+                           nparts: int, C: int):
+    """(dbeta, dgamma) from dgrad-epilogue-fused reduce partials."""
+    lib = require_lib()
+    dev = part_db.device
+    if nparts > 512:
+        G = max(8, 512 // max(1, (C + 31) // 32))
+        stage = torch.empty(2, G, C, dtype=torch.float32, device=dev)
+        check(lib.ddlw_bn_parts_fold(_p(part_db), _p(part_dg), _p(stage[0]),
+                                     _p(stage[1]), ctypes.c_long(nparts),
+                                     ctypes.c_int(C), ctypes.c_int(G),
+                                     ctypes.c_void_p(current_stream_ptr())),
+              "bn_parts_fold")
+        part_db, part_dg, nparts = stage[0], stage[1], G
+    dbeta = torch.empty(C, dtype=torch.float32, device=dev)
+    dgamma = torch.empty(C, dtype=torch.float32, device=dev)
+    check(lib.ddlw_bn_grad_finalize_n(_p(part_db), _p(part_dg), _p(dbeta),
+                                      _p(dgamma), ctypes.c_int(C),
+                                      ctypes.c_int(nparts),
+                                      ctypes.c_void_p(current_stream_ptr())),
+          "bn_grad_finalize_n")
+    return dbeta, dgamma
+
+
 def bn_apply(x: torch.Tensor, res: Optional[torch.Tensor], mean, rstd, gamma, beta,
              relu: bool):
     """Returns (y, mask): mask is a uint8 tensor of rows*C/8 relu bits (one

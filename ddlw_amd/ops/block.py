@@ -68,6 +68,28 @@ def _fwd_conv(x, w, stride, padding, hip_fwd):
     return torch.nn.functional.conv2d(x, w.to(x.dtype), None, stride, padding)
 
 
+def _dgrad_bn_reduce(dy, conv_in, weight, stride, padding, hip_dgrad,
+                     x_bn, mask, mean, rstd):
+    """conv dgrad + the NEXT BatchNorm backward's (dbeta, dgamma) reduce.
+    On the stride-1 hip route the dgrad epilogue emits the reduce partials
+    (dy never re-read); otherwise dgrad then the standalone reduce pass.
+    DDLW_FUSED_BNB=0 disables (A/B lever). Returns (dx, dbeta, dgamma)."""
+    st = stride[0] if isinstance(stride, (tuple, list)) else stride
+    if (hip_dgrad and st == 1
+            and os.environ.get("DDLW_FUSED_BNB", "1") == "1"):
+        dx, parts, np_ = conv_gemm.conv_dgrad_kernel(
+            dy, weight.to(torch.bfloat16), conv_in.shape, padding, stride,
+            bnb=(x_bn, mask, mean, rstd))
+        db, dg = binding.bn_grad_finalize_parts(
+            parts[0], parts[1], np_, x_bn.shape[1])
+        return dx, db, dg
+    dx, _ = conv_gemm.conv_backward(dy, conv_in, weight, stride, padding,
+                                    hip_dgrad, False, need_dw=False)
+    db, dg = binding.bn_bwd_reduce(dx, mask, x_bn, mean, rstd,
+                                   mask is not None)
+    return dx, db, dg
+
+
 def _fwd_conv_bn_stats(x, w, stride, padding, hip_fwd, bn):
     """conv + BN batch statistics. On the hip path the conv epilogue emits
     per-tile partial sums (no separate k_bn_stats read of the conv output);
@@ -181,17 +203,13 @@ class _BottleneckFn(torch.autograd.Function):
         dt3, dres = binding.bn_bwd_dx(dy, mask3, t3, m3, s3, g3, db3, dg3,
                                       True, True)
         dw3 = wgrad(dt3, a2, w3, 1, 0, r3g)
-        da2, _ = conv_gemm.conv_backward(dt3, a2, w3, 1, 0, r3d, False,
-                                         need_dw=False)
-
-        db2, dg2 = binding.bn_bwd_reduce(da2, mask2, t2, m2, s2, True)
+        da2, db2, dg2 = _dgrad_bn_reduce(dt3, a2, w3, 1, 0, r3d,
+                                         t2, mask2, m2, s2)
         dt2, _ = binding.bn_bwd_dx(da2, mask2, t2, m2, s2, g2, db2, dg2,
                                    True, False)
         dw2 = wgrad(dt2, a1, w2, stride, 1, r2g)
-        da1, _ = conv_gemm.conv_backward(dt2, a1, w2, stride, 1, r2d, False,
-                                         need_dw=False)
-
-        db1, dg1 = binding.bn_bwd_reduce(da1, mask1, t1, m1, s1, True)
+        da1, db1, dg1 = _dgrad_bn_reduce(dt2, a1, w2, stride, 1, r2d,
+                                         t1, mask1, m1, s1)
         dt1, _ = binding.bn_bwd_dx(da1, mask1, t1, m1, s1, g1, db1, dg1,
                                    True, False)
 

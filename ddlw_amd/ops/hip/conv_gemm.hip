@@ -102,6 +102,82 @@ __device__ __forceinline__ void bn_partials_16(
   }
 }
 
+// BN-BACKWARD reduce fusion: a conv dgrad's epilogue already holds the dy
+// of the NEXT BatchNorm backward in registers — accumulating the masked
+// (dbeta, dgamma) partials here removes k_bn_bwd_reduce's re-read of dy
+// (SURVEY.md §2.4 fused-op mandate; reduce was ~9% of the train step).
+//   part_db = sum(mask * v),  part_dg = sum(mask * v * (x - mean) * rstd)
+template <int MF, int NF, int BM, int BN, int WM, int WN, typename ACC,
+          typename OutRow>
+__device__ __forceinline__ void bnb_partials_16(
+    const ACC& acc, const bf16_t* __restrict__ accp,
+    const bf16_t* __restrict__ bx, const unsigned char* __restrict__ bmask,
+    const float* __restrict__ bmean, const float* __restrict__ brstd,
+    float* __restrict__ p_db, float* __restrict__ p_dg,
+    long tile_m, int tile_n, int wr, int wc, long M, int K, OutRow out_row) {
+  const int lane = threadIdx.x & 63;
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+  float s[NF], q[NF], mn[NF], rs[NF];
+  #pragma unroll
+  for (int ni = 0; ni < NF; ++ni) {
+    int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+    s[ni] = 0.f;
+    q[ni] = 0.f;
+    mn[ni] = (j < K) ? bmean[j] : 0.f;
+    rs[ni] = (j < K) ? brstd[j] : 0.f;
+  }
+  #pragma unroll
+  for (int mi = 0; mi < MF; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
+      int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+      if (j >= K) continue;
+      #pragma unroll
+      for (int qq = 0; qq < 4; ++qq) {
+        long m = tile_m * BM + (long)wr * WM + mi * 16 + d_row0 + qq;
+        if (m < M) {
+          long oi = out_row(m) * K + j;
+          float v = acc[mi][ni][qq];
+          if (accp) {
+            union { unsigned i; float f; } ca;
+            ca.i = (unsigned)accp[oi] << 16;
+            v += ca.f;
+          }
+          union { unsigned i; float f; } cv;
+          cv.i = (unsigned)f2b_hw(v) << 16;  // the stored bf16 value
+          bool on = true;
+          if (bmask)
+            on = (bmask[oi >> 3] >> (j & 7)) & 1;
+          if (on) {
+            union { unsigned i; float f; } cx;
+            cx.i = (unsigned)bx[oi] << 16;
+            s[ni] += cv.f;
+            q[ni] += cv.f * (cx.f - mn[ni]) * rs[ni];
+          }
+        }
+      }
+    }
+  #pragma unroll
+  for (int ni = 0; ni < NF; ++ni) {
+    s[ni] += __shfl_xor(s[ni], 16, 64);
+    s[ni] += __shfl_xor(s[ni], 32, 64);
+    q[ni] += __shfl_xor(q[ni], 16, 64);
+    q[ni] += __shfl_xor(q[ni], 32, 64);
+  }
+  if (lane < 16) {
+    const long prow_ = tile_m * (BM / WM) + wr;
+    #pragma unroll
+    for (int ni = 0; ni < NF; ++ni) {
+      int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+      if (j < K) {
+        p_db[prow_ * K + j] = s[ni];
+        p_dg[prow_ * K + j] = q[ni];
+      }
+    }
+  }
+}
+
 template <int BM, int BN, bool EPI_LDS, int BUFS, bool M32EN = true>
 __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
@@ -113,7 +189,11 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     unsigned long long magic_ho, unsigned shift_ho,
     const bf16_t* __restrict__ accp,   // optional epilogue accumulate input
     float* __restrict__ bn_ps,         // optional BN partial sums [nparts][K]
-    float* __restrict__ bn_pq) {       // optional BN partial sumsq
+    float* __restrict__ bn_pq,         // optional BN partial sumsq
+    const bf16_t* __restrict__ bnb_x,  // != null: BN-BWD mode — partials
+    const unsigned char* __restrict__ bnb_mask,  // become (dbeta, dgamma)
+    const float* __restrict__ bnb_mean,
+    const float* __restrict__ bnb_rstd) {
   // accp != nullptr: y = conv + accp (read at the output index). Used to
   // fuse the residual-join gradient add (d_block_input = conv1_dgrad +
   // d_identity) into the dgrad epilogue — saves the engine's separate
@@ -390,7 +470,15 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           }
         }
       }
-      if (bn_ps) bn_partials_16<MF, NF, BM, BN, WM, WN>(acc, bn_ps, bn_pq, tile_m, tile_n, wr, wc, M, K);
+      if (bn_ps) {
+        if (bnb_x)
+          bnb_partials_16<MF, NF, BM, BN, WM, WN>(
+              acc, accp, bnb_x, bnb_mask, bnb_mean, bnb_rstd, bn_ps, bn_pq,
+              tile_m, tile_n, wr, wc, M, K, out_row);
+        else
+          bn_partials_16<MF, NF, BM, BN, WM, WN>(acc, bn_ps, bn_pq, tile_m,
+                                                 tile_n, wr, wc, M, K);
+      }
     }
     return;
   }
@@ -424,9 +512,17 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   const int e_ch = lane % CPL;       // chunk within the row
   const long m_base = tile_m * BM + wr * WM;
   const int j_base = tile_n * BN + wc * WN + e_ch * 8;
-  float s8[8], q8[8];
+  float s8[8], q8[8], bnm8[8], bnr8[8];
   #pragma unroll
   for (int e = 0; e < 8; ++e) { s8[e] = 0.f; q8[e] = 0.f; }
+  if (bn_ps && bnb_x) {
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int j = j_base + e;
+      bnm8[e] = (j < K) ? bnb_mean[j] : 0.f;
+      bnr8[e] = (j < K) ? bnb_rstd[j] : 0.f;
+    }
+  }
   #pragma unroll
   for (int it = 0; it < WM / RPI; ++it) {
     const int row = it * RPI + e_row;
@@ -452,13 +548,31 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
         }
         *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
         if (bn_ps) {
-          // fused BN partials (pure VALU; never combined with accp)
-          #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
-            float f = b2f((bf16_t)(wd >> ((e & 1) * 16)));
-            s8[e] += f;
-            q8[e] += f * f;
+          if (bnb_x) {
+            // BN-BWD partials: masked dy and dy*xhat (j_base is 8-aligned
+            // so the mask byte covers exactly this chunk)
+            unsigned mb = bnb_mask ? bnb_mask[(orow * K + j_base) >> 3] : 0xffu;
+            uint4 xv = *reinterpret_cast<const uint4*>(bnb_x + orow * K + j_base);
+            #pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
+              unsigned xw = (e < 2) ? xv.x : (e < 4) ? xv.y : (e < 6) ? xv.z : xv.w;
+              if ((mb >> e) & 1) {
+                float f = b2f((bf16_t)(wd >> ((e & 1) * 16)));
+                float xf = b2f((bf16_t)(xw >> ((e & 1) * 16)));
+                s8[e] += f;
+                q8[e] += f * (xf - bnm8[e]) * bnr8[e];
+              }
+            }
+          } else {
+            // fused fwd BN statistics
+            #pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
+              float f = b2f((bf16_t)(wd >> ((e & 1) * 16)));
+              s8[e] += f;
+              q8[e] += f * f;
+            }
           }
         }
       } else {
@@ -474,8 +588,18 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
             y[orow * K + j_base + e] = ov;
             if (bn_ps) {
               float f = b2f(ov);
-              s8[e] += f;
-              q8[e] += f * f;
+              if (bnb_x) {
+                bool on = !bnb_mask ||
+                          ((bnb_mask[(orow * K + j_base) >> 3] >> e) & 1);
+                if (on) {
+                  s8[e] += f;
+                  q8[e] += f * (b2f(bnb_x[orow * K + j_base + e]) - bnm8[e]) *
+                           bnr8[e];
+                }
+              } else {
+                s8[e] += f;
+                q8[e] += f * f;
+              }
             }
           }
         }
@@ -531,7 +655,11 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     unsigned long long magic_wo, unsigned shift_wo,
     unsigned long long magic_ho, unsigned shift_ho,
     const bf16_t* __restrict__ accp,
-    float* __restrict__ bn_ps, float* __restrict__ bn_pq) {
+    float* __restrict__ bn_ps, float* __restrict__ bn_pq,
+    const bf16_t* __restrict__ bnb_x,
+    const unsigned char* __restrict__ bnb_mask,
+    const float* __restrict__ bnb_mean,
+    const float* __restrict__ bnb_rstd) {
   constexpr int BK = 64;
   constexpr int WAVES = 8;            // 4 (M) x 2 (N)
   constexpr int WM = BM / 4, WN = BN / 2;
@@ -751,7 +879,15 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm_wide(
     epilogue(std::true_type{});
   else
     epilogue(std::false_type{});
-  if (bn_ps) bn_partials_16<MF, NF, BM, BN, WM, WN>(acc, bn_ps, bn_pq, tile_m, tile_n, wr, wc, M, K);
+  if (bn_ps) {
+    if (bnb_x)
+      bnb_partials_16<MF, NF, BM, BN, WM, WN>(
+          acc, accp, bnb_x, bnb_mask, bnb_mean, bnb_rstd, bn_ps, bn_pq,
+          tile_m, tile_n, wr, wc, M, K, out_row);
+    else
+      bn_partials_16<MF, NF, BM, BN, WM, WN>(acc, bn_ps, bn_pq, tile_m,
+                                             tile_n, wr, wc, M, K);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -822,6 +958,8 @@ DDLW_EXPORT long ddlw_conv_fwd_nparts(int N, int C, int K, int Ho, int Wo,
 static int conv_fwd_launch(const void* x, const void* w, void* y,
                            const void* zpage, const void* acc,
                            void* bn_ps, void* bn_pq,
+                           const void* bnb_x, const void* bnb_mask,
+                           const void* bnb_mean, const void* bnb_rstd,
                            int N, int H, int W_, int C, int K,
                            int Ho, int Wo, int R, int S, int stride,
                            int pad, int oH, int oW, int oS, void* stream) {
@@ -850,7 +988,9 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
                        (bf16_t*)y, (const bf16_t*)zpage, N, H, W_, C, K, Ho,   \
                        Wo, R, S, stride, pad, (int)grid, oH, oW, oS, mg_wo,    \
                        sh_wo, mg_ho, sh_ho, (const bf16_t*)acc,                \
-                       (float*)bn_ps, (float*)bn_pq)
+                       (float*)bn_ps, (float*)bn_pq, (const bf16_t*)bnb_x,     \
+                       (const unsigned char*)bnb_mask,                         \
+                       (const float*)bnb_mean, (const float*)bnb_rstd)
     if (p.bn == 256)
       WIDE_LAUNCH(256, 2);
     else
@@ -867,14 +1007,20 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
                          (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
                          N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
                          (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho,   \
-                         (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq);   \
+                         (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq,    \
+                         (const bf16_t*)bnb_x,                                \
+                         (const unsigned char*)bnb_mask,                      \
+                         (const float*)bnb_mean, (const float*)bnb_rstd);     \
     else                                                                      \
       hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, true>),         \
                          dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
                          (const bf16_t*)w, (bf16_t*)y, (const bf16_t*)zpage,  \
                          N, H, W_, C, K, Ho, Wo, R, S, stride, pad,           \
                          (int)grid, oH, oW, oS, mg_wo, sh_wo, mg_ho, sh_ho,   \
-                         (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq);   \
+                         (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq,    \
+                         (const bf16_t*)bnb_x,                                \
+                         (const unsigned char*)bnb_mask,                      \
+                         (const float*)bnb_mean, (const float*)bnb_rstd);     \
   } while (0)
   if (p.bn == 128) {
     if (p.epi_lds) {
@@ -911,8 +1057,9 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
                                         int Ho, int Wo, int R, int S, int stride,
                                         int pad, int oH, int oW, int oS,
                                         void* stream) {
-  return conv_fwd_launch(x, w, y, zpage, acc, nullptr, nullptr, N, H, W_, C,
-                         K, Ho, Wo, R, S, stride, pad, oH, oW, oS, stream);
+  return conv_fwd_launch(x, w, y, zpage, acc, nullptr, nullptr, nullptr,
+                         nullptr, nullptr, nullptr, N, H, W_, C, K, Ho, Wo, R,
+                         S, stride, pad, oH, oW, oS, stream);
 }
 
 DDLW_EXPORT int ddlw_conv_fwd_igemm_stats(const void* x, const void* w, void* y,
@@ -921,8 +1068,23 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_stats(const void* x, const void* w, void* y,
                                           int N, int H, int W_, int C, int K,
                                           int Ho, int Wo, int R, int S,
                                           int stride, int pad, void* stream) {
-  return conv_fwd_launch(x, w, y, zpage, nullptr, bn_ps, bn_pq, N, H, W_, C,
-                         K, Ho, Wo, R, S, stride, pad, Ho, Wo, 1, stream);
+  return conv_fwd_launch(x, w, y, zpage, nullptr, bn_ps, bn_pq, nullptr,
+                         nullptr, nullptr, nullptr, N, H, W_, C, K, Ho, Wo, R,
+                         S, stride, pad, Ho, Wo, 1, stream);
+}
+
+// dgrad + fused BN-backward reduce: y = dgrad output (the BN's dy); the
+// epilogue also emits (dbeta, dgamma) partials against (bnb_x, mask,
+// mean, rstd) so k_bn_bwd_reduce never re-reads dy.
+DDLW_EXPORT int ddlw_conv_fwd_igemm_bnb(
+    const void* x, const void* w, void* y, const void* zpage, const void* acc,
+    const void* bnb_x, const void* bnb_mask, const void* bnb_mean,
+    const void* bnb_rstd, void* p_db, void* p_dg,
+    int N, int H, int W_, int C, int K, int Ho, int Wo, int R, int S,
+    int stride, int pad, void* stream) {
+  return conv_fwd_launch(x, w, y, zpage, acc, p_db, p_dg, bnb_x, bnb_mask,
+                         bnb_mean, bnb_rstd, N, H, W_, C, K, Ho, Wo, R, S,
+                         stride, pad, Ho, Wo, 1, stream);
 }
 
 DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,

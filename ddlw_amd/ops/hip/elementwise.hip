@@ -839,6 +839,17 @@ DDLW_EXPORT int ddlw_bn_grad_finalize(const void* part_db, const void* part_dg,
   DDLW_CHECK_LAUNCH();
 }
 
+DDLW_EXPORT int ddlw_bn_grad_finalize_n(const void* part_db,
+                                        const void* part_dg, void* dbeta,
+                                        void* dgamma, int C, int nparts,
+                                        void* stream) {
+  hipLaunchKernelGGL(k_bn_grad_finalize, dim3((C + 31) / 32), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)part_db,
+                     (const float*)part_dg, (float*)dbeta, (float*)dgamma, C,
+                     nparts);
+  DDLW_CHECK_LAUNCH();
+}
+
 DDLW_EXPORT int ddlw_bn_bwd_dx(const void* dy, const void* mask, const void* x,
                                const void* mean, const void* rstd,
                                const void* gamma, const void* dbeta,

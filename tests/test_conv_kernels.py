@@ -340,3 +340,29 @@ def test_stem_conv_autograd_route():
     y.float().square().mean().backward()
     assert conv.weight.grad is not None
     assert torch.isfinite(conv.weight.grad.float()).all()
+
+
+def test_conv_dgrad_fused_bnb_parity():
+    """dgrad + fused BN-backward reduce vs the standalone pair."""
+    from ddlw_amd.ops import binding, conv_gemm
+
+    torch.manual_seed(6)
+    for C, K, R, pad in ((128, 256, 1, 0), (128, 128, 3, 1)):
+        B, H = 9, 14
+        w = _cl(torch.randn(K, C, R, R, device=_cuda()).to(torch.bfloat16) * 0.1)
+        dy = _cl(torch.randn(B, K, H, H, device=_cuda()).to(torch.bfloat16))
+        x_bn = _cl(torch.randn(B, C, H, H, device=_cuda()).to(torch.bfloat16))
+        mean = torch.randn(C, device=_cuda())
+        rstd = torch.rand(C, device=_cuda()) + 0.5
+        mask = torch.randint(0, 256, (B * H * H * (C // 8),),
+                             dtype=torch.uint8, device=_cuda())
+        dx_f, parts, np_ = conv_gemm.conv_dgrad_kernel(
+            dy, w, (B, C, H, H), pad, 1, bnb=(x_bn, mask, mean, rstd))
+        db_f, dg_f = binding.bn_grad_finalize_parts(parts[0], parts[1], np_, C)
+        dx_u = conv_gemm.conv_dgrad_kernel(dy, w, (B, C, H, H), pad, 1)
+        db_u, dg_u = binding.bn_bwd_reduce(dx_u, mask, x_bn, mean, rstd, True)
+        assert torch.equal(dx_f, dx_u), (R, pad)
+        sb = db_u.abs().max() + 1e-3
+        sg = dg_u.abs().max() + 1e-3
+        assert ((db_f - db_u).abs().max() / sb).item() < 1e-3, (R, pad)
+        assert ((dg_f - dg_u).abs().max() / sg).item() < 1e-3, (R, pad)
