@@ -913,7 +913,8 @@ struct TilePick {
 // One tile decision for launcher + nparts query (must stay in sync by
 // construction). want_stats forces the 16x16 direct epilogue (the only
 // ones carrying the bn-partials code).
-static TilePick pick_tile(long M, int K, int C, int T, bool want_stats) {
+static TilePick pick_tile(long M, int K, int C, int T, bool want_stats,
+                          bool want_bnb = false) {
   int mfma_pref, wide_pref;
   {
     const char* e = getenv("DDLW_CONV_MFMA");
@@ -922,7 +923,7 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats) {
     wide_pref = e2 ? (e2[0] == '0' ? 0 : 1) : -1;  // -1 = auto
   }
   TilePick p{};
-  p.m32 = (mfma_pref == 32) && !want_stats;
+  p.m32 = (mfma_pref == 32) && !want_stats && !want_bnb;
   // measured routing (bench/tools/wide_check.py on MI355X): the 256x256
   // 2-buf wide kernel wins (+20-28%) where K >= 256, the K-loop is deep
   // enough to amortize its prologue (T >= 9), and the grid still fills
@@ -931,7 +932,8 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats) {
   // auto route (kept for DDLW_CONV_WIDE=1 A/Bs).
   const long wide_blocks = cdiv(M, 256) * cdiv(K, 256);
   const bool wide_auto = (K >= 256) && (T >= 9) && (wide_blocks >= 150);
-  if ((wide_pref == 1 && K >= 128 && T >= 4) || (wide_pref == -1 && wide_auto)) {
+  if (!want_bnb &&
+      ((wide_pref == 1 && K >= 128 && T >= 4) || (wide_pref == -1 && wide_auto))) {
     p.wide = true;
     p.bm = 256;
     p.bn = (K >= 256) ? 256 : 128;
@@ -942,16 +944,19 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats) {
   p.bm = 128;
   p.wvm = 2;
   p.bn = (K >= 128) ? 128 : (K >= 64 ? 64 : 32);
-  p.epi_lds = (T <= 4);
+  // bnb (fused BN-backward reduce) must use the LDS-bounce epilogue on
+  // EVERY shape: the direct epilogue's per-element stride-K x loads are
+  // uncoalesced scalar L2 round trips (measured -7% whole-model)
+  p.epi_lds = (T <= 4) || want_bnb;
   p.bufs = (T == 1) ? 1 : 2;
   return p;
 }
 
 DDLW_EXPORT long ddlw_conv_fwd_nparts(int N, int C, int K, int Ho, int Wo,
-                                      int R, int S) {
+                                      int R, int S, int bnb) {
   long M = (long)N * Ho * Wo;
   const int T = R * S * (C / 64);
-  TilePick p = pick_tile(M, K, C, T, true);
+  TilePick p = pick_tile(M, K, C, T, true, bnb != 0);
   return cdiv(M, p.bm) * p.wvm;
 }
 
@@ -979,7 +984,8 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
   const int T = R * S * (C / 64);
   hipStream_t st = (hipStream_t)stream;
   const bool want_stats = bn_ps != nullptr;
-  TilePick p = pick_tile(M, K, C, T, want_stats);
+  const bool want_bnb = bnb_x != nullptr;
+  TilePick p = pick_tile(M, K, C, T, want_stats, want_bnb);
   if (p.wide) {
     long grid = cdiv(M, 256) * cdiv(K, p.bn);
 #define WIDE_LAUNCH(BN_, NB_)                                                  \
