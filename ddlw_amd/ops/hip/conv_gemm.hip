@@ -924,14 +924,14 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats,
   }
   TilePick p{};
   p.m32 = (mfma_pref == 32) && !want_stats && !want_bnb;
-  // measured routing (bench/tools/wide_check.py on MI355X): the 256x256
-  // 2-buf wide kernel wins (+20-28%) where K >= 256, the K-loop is deep
-  // enough to amortize its prologue (T >= 9), and the grid still fills
-  // most CUs (>= 150 blocks); everywhere else the 128x128 2-block/CU
-  // kernel wins on occupancy. The 256x128 3-buf variant never won an
-  // auto route (kept for DDLW_CONV_WIDE=1 A/Bs).
-  const long wide_blocks = cdiv(M, 256) * cdiv(K, 256);
-  const bool wide_auto = (K >= 256) && (T >= 9) && (wide_blocks >= 150);
+  // measured (bench/tools/wide_check.py + whole-model A/B on MI355X): the
+  // 256x256 2-buf wide kernel wins +20-28% on its isolated shapes
+  // (K >= 256, T >= 9, >= 150 blocks) but the full ResNet-50 step runs
+  // ~1% SLOWER with it auto-routed (repeated A/B; DVFS/L2-context effect:
+  // the denser kernels depress the whole-step clock by more than their
+  // isolated win). Auto routing is therefore OFF — a documented negative
+  // result; DDLW_CONV_WIDE=1 forces it for probes.
+  const bool wide_auto = false;
   if (!want_bnb &&
       ((wide_pref == 1 && K >= 128 && T >= 4) || (wide_pref == -1 && wide_auto))) {
     p.wide = true;
