@@ -43,9 +43,8 @@ __device__ __forceinline__ unsigned lds_u32(const bf16_t* p) {
   return (unsigned)(unsigned long long)(const LDS_AS bf16_t*)p;
 }
 
-#define WG_BM 64  // m rows per k-step
-
-template <int BK, int BC>  // output tile: BK x BC (k x c)
+template <int BK, int BC, int WG_BM = 64>  // output tile: BK x BC (k x c);
+                                           // WG_BM = m rows per k-step
 __global__ __launch_bounds__(256) void k_conv_wgrad(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ x,
     const bf16_t* __restrict__ zpage,
@@ -57,8 +56,9 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
     unsigned long long magic_ho, unsigned shift_ho) {
   constexpr int WK = BK / 2, WC = BC / 2;   // per-wave tile (2x2 wave grid)
   constexpr int KF = WK / 16, CF = WC / 16; // fragments
-  constexpr int PA = BK / 16 * 2;           // 1-KiB dy pieces (k16 x m32)
-  constexpr int PB = BC / 16 * 2;           // 1-KiB x pieces
+  constexpr int MG = WG_BM / 32;            // 32-m groups per k-step
+  constexpr int PA = BK / 16 * MG;          // 1-KiB dy pieces (k16 x m32)
+  constexpr int PB = BC / 16 * MG;          // 1-KiB x pieces
   constexpr int TILE = WG_BM * BK;          // elements per dy tile
   constexpr int BUF = WG_BM * (BK + BC);
   __shared__ __attribute__((aligned(16))) bf16_t smem[2 * BUF];
@@ -91,13 +91,13 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
   const int s_koff = s_half * 8;               // k offset within the k16
 
   // per-(m32) step state: global row pointers + validity for this lane
-  const bf16_t* dyrow[2];
-  const bf16_t* xrow[2];
-  bool dv[2], xv[2];
+  const bf16_t* dyrow[MG];
+  const bf16_t* xrow[MG];
+  bool dv[MG], xv[MG];
 
   auto decompose = [&](long mbase) {
     #pragma unroll
-    for (int g = 0; g < 2; ++g) {
+    for (int g = 0; g < MG; ++g) {
       long m = mbase + g * 32 + s_mlocal;
       dv[g] = m < m1;
       dyrow[g] = dv[g] ? (dy + m * K) : zpage;
@@ -126,7 +126,7 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
     bf16_t* lx = ldy + TILE;           // x tile image
     #pragma unroll
     for (int p = wave; p < PA; p += 4) {
-      const int k16 = p >> 1, g = p & 1;
+      const int k16 = p / MG, g = p % MG;
       int kk = tile_k * BK + k16 * 16 + s_koff;
       const bf16_t* src = (dv[g] && kk < K) ? (dyrow[g] + kk) : zpage;
       __builtin_amdgcn_global_load_lds(
@@ -134,7 +134,7 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
     }
     #pragma unroll
     for (int p = wave; p < PB; p += 4) {
-      const int c16 = p >> 1, g = p & 1;
+      const int c16 = p / MG, g = p % MG;
       int cc = tile_c * BC + c16 * 16 + s_koff;
       const bf16_t* src = (xv[g] && cc < C) ? (xrow[g] + cc) : zpage;
       __builtin_amdgcn_global_load_lds(
@@ -163,7 +163,7 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
     bf16_t* ldy = smem + cur * BUF;
     bf16_t* lx = ldy + TILE;
     #pragma unroll
-    for (int mh = 0; mh < 2; ++mh) {  // two 32-m halves of the 64-m step
+    for (int mh = 0; mh < MG; ++mh) {  // 32-m groups of the WG_BM-m step
       // fragment loads: 2 hardware transpose reads per fragment
       tr64_t fk0[KF], fk1[KF], fc0[CF], fc1[CF];
       // per-lane address = base + lane*8 B: each 16-lane subgroup reads one
@@ -171,7 +171,7 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
       #pragma unroll
       for (int a = 0; a < KF; ++a) {
         const int k16 = (wr * WK) / 16 + a;
-        unsigned addr = lds_u32(ldy + ((k16 * 2 + mh) * 8) * 64) + lane * 8;
+        unsigned addr = lds_u32(ldy + ((k16 * MG + mh) * 8) * 64) + lane * 8;
         asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(fk0[a]) : "v"(addr));
         asm volatile("ds_read_b64_tr_b16 %0, %1 offset:512"
                      : "=v"(fk1[a]) : "v"(addr));
@@ -179,7 +179,7 @@ __global__ __launch_bounds__(256) void k_conv_wgrad(
       #pragma unroll
       for (int b = 0; b < CF; ++b) {
         const int c16 = (wc2 * WC) / 16 + b;
-        unsigned addr = lds_u32(lx + ((c16 * 2 + mh) * 8) * 64) + lane * 8;
+        unsigned addr = lds_u32(lx + ((c16 * MG + mh) * 8) * 64) + lane * 8;
         asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(fc0[b]) : "v"(addr));
         asm volatile("ds_read_b64_tr_b16 %0, %1 offset:512"
                      : "=v"(fc1[b]) : "v"(addr));
@@ -279,6 +279,15 @@ DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, const void* zpage
   wg_magic((unsigned)Ho, &mg_ho, &sh_ho);
   long m_per_split = wg_cdiv(M, split);
   hipStream_t st = (hipStream_t)stream;
+#define WLAUNCH3(BK, BC, BM_)                                                 \
+  do {                                                                        \
+    dim3 grid((int)(wg_cdiv(K, BK) * wg_cdiv(C, BC)), R * S, split);          \
+    hipLaunchKernelGGL((k_conv_wgrad<BK, BC, BM_>), grid, dim3(256), 0, st,   \
+                       (const bf16_t*)dy, (const bf16_t*)x,                   \
+                       (const bf16_t*)zpage, (float*)slab, N, H, W_, C, K,    \
+                       Ho, Wo, R, S, stride, pad, split, m_per_split, mg_wo,  \
+                       sh_wo, mg_ho, sh_ho);                                  \
+  } while (0)
 #define WLAUNCH(BK, BC)                                                       \
   do {                                                                        \
     dim3 grid((int)(wg_cdiv(K, BK) * wg_cdiv(C, BC)), R * S, split);          \
@@ -288,11 +297,20 @@ DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, const void* zpage
                        Ho, Wo, R, S, stride, pad, split, m_per_split, mg_wo,  \
                        sh_wo, mg_ho, sh_ho);                                  \
   } while (0)
+  int wg_bm = 64;
+  {
+    const char* e = getenv("DDLW_WGRAD_BM");
+    if (e && e[0] == '1') wg_bm = 128;       // "128"
+    else if (e && e[0] == '6') wg_bm = 64;   // "64"
+    else if (K < 128 && C < 128) wg_bm = 128;  // default: deeper m-step on
+  }                                            // the small 64x64 tiles
   if (K >= 128 && C >= 128) WLAUNCH(128, 128);
   else if (C >= 128) WLAUNCH(64, 128);
   else if (K >= 128) WLAUNCH(128, 64);
+  else if (wg_bm == 128) WLAUNCH3(64, 64, 128);
   else WLAUNCH(64, 64);
 #undef WLAUNCH
+#undef WLAUNCH3
   {
     hipError_t err_ = hipGetLastError();
     if (err_ != hipSuccess) { ddlw_set_error(hipGetErrorString(err_)); return 1; }
