@@ -301,9 +301,11 @@ DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, const void* zpage
   {
     const char* e = getenv("DDLW_WGRAD_BM");
     if (e && e[0] == '1') wg_bm = 128;       // "128"
-    else if (e && e[0] == '6') wg_bm = 64;   // "64"
-    else if (K < 128 && C < 128) wg_bm = 128;  // default: deeper m-step on
-  }                                            // the small 64x64 tiles
+    // default stays 64: the deeper 128-m k-step measured EQUAL OR SLOWER
+    // on every 64x64-tile route (110-234 vs 120-248 TF) — the small-tile
+    // wgrad is staging-bandwidth-bound (AI = 32 flops/B), not barrier- or
+    // pipeline-depth-bound. Documented negative result; env keeps the A/B.
+  }
   if (K >= 128 && C >= 128) WLAUNCH(128, 128);
   else if (C >= 128) WLAUNCH(64, 128);
   else if (K >= 128) WLAUNCH(128, 64);
