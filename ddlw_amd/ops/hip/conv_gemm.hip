@@ -178,6 +178,20 @@ __device__ __forceinline__ void bnb_partials_16(
   }
 }
 
+// stride-2 dgrad, merged parity classes: blockIdx.y picks one of four
+// (h%2, w%2) output-parity classes, each a small stride-1 conv of dy with
+// its own gathered taps and scattered output offset. One launch keeps the
+// short-K classes (1 tap) resident WITH the 4-tap class instead of four
+// serialized latency-bound launches.
+struct S2Class {
+  long woff;   // offset into the concatenated per-class weights
+  long yoff;   // output element offset ((dh*iw + dw) * C)
+  int R, S;    // taps of this class
+};
+struct S2Quad {
+  S2Class c[4];
+};
+
 template <int BM, int BN, bool EPI_LDS, int BUFS, bool M32EN = true>
 __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
@@ -193,7 +207,15 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const bf16_t* __restrict__ bnb_x,  // != null: BN-BWD mode — partials
     const unsigned char* __restrict__ bnb_mask,  // become (dbeta, dgamma)
     const float* __restrict__ bnb_mean,
-    const float* __restrict__ bnb_rstd) {
+    const float* __restrict__ bnb_rstd,
+    int s2_merged = 0, S2Quad quad = {}) {
+  if (s2_merged) {
+    const S2Class& cc = quad.c[blockIdx.y];
+    w += cc.woff;
+    y += cc.yoff;
+    R = cc.R;
+    S = cc.S;
+  }
   // accp != nullptr: y = conv + accp (read at the output index). Used to
   // fuse the residual-join gradient add (d_block_input = conv1_dgrad +
   // d_identity) into the dgrad epilogue — saves the engine's separate
@@ -1088,6 +1110,49 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_bnb(
   return conv_fwd_launch(x, w, y, zpage, acc, p_db, p_dg, bnb_x, bnb_mask,
                          bnb_mean, bnb_rstd, N, H, W_, C, K, Ho, Wo, R, S,
                          stride, pad, Ho, Wo, 1, stream);
+}
+
+// merged-class stride-2 3x3 dgrad (even ih/iw): dx = scattered union of 4
+// parity-class stride-1 convs of dy, one launch (grid.y = class).
+DDLW_EXPORT int ddlw_conv_dgrad_s2m(const void* dy, const void* wcat, void* dx,
+                                    const void* zpage, int N, int Ho, int Wo,
+                                    int Kdy, int Cdx, int oh, int ow, int ih,
+                                    int iw, const long* woffs,
+                                    const long* yoffs, const int* crs,
+                                    void* stream) {
+  if (Kdy % 64 != 0) {
+    ddlw_set_error("conv_dgrad_s2m: dy channels must be a multiple of 64");
+    return 2;
+  }
+  long M = (long)N * oh * ow;
+  if (M >= (1ll << 31)) {
+    ddlw_set_error("conv_dgrad_s2m: M >= 2^31 unsupported");
+    return 2;
+  }
+  unsigned long long mg_wo, mg_ho;
+  unsigned sh_wo, sh_ho;
+  make_magic((unsigned)ow, &mg_wo, &sh_wo);
+  make_magic((unsigned)oh, &mg_ho, &sh_ho);
+  S2Quad q;
+  for (int i = 0; i < 4; ++i)
+    q.c[i] = S2Class{woffs[i], yoffs[i], crs[2 * i], crs[2 * i + 1]};
+  hipStream_t st = (hipStream_t)stream;
+#define S2LAUNCH(BN_)                                                          \
+  do {                                                                         \
+    long gx = cdiv(M, 128) * cdiv(Cdx, BN_);                                   \
+    hipLaunchKernelGGL((k_conv_fwd_igemm<128, BN_, true, 2, false>),           \
+                       dim3((int)gx, 4), dim3(256), 0, st, (const bf16_t*)dy,  \
+                       (const bf16_t*)wcat, (bf16_t*)dx, (const bf16_t*)zpage, \
+                       N, Ho, Wo, Kdy, Cdx, oh, ow, 2, 2, 1, 0, (int)gx, ih,   \
+                       iw, 2, mg_wo, sh_wo, mg_ho, sh_ho, nullptr, nullptr,    \
+                       nullptr, nullptr, nullptr, nullptr, nullptr, 1, q);     \
+  } while (0)
+  if (Cdx >= 128)
+    S2LAUNCH(128);
+  else
+    S2LAUNCH(64);
+#undef S2LAUNCH
+  DDLW_CHECK_LAUNCH();
 }
 
 DDLW_EXPORT int ddlw_conv_fwd_igemm(const void* x, const void* w, void* y,
