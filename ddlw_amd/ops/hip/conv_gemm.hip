@@ -1137,10 +1137,24 @@ DDLW_EXPORT int ddlw_conv_dgrad_s2m(const void* dy, const void* wcat, void* dx,
   for (int i = 0; i < 4; ++i)
     q.c[i] = S2Class{woffs[i], yoffs[i], crs[2 * i], crs[2 * i + 1]};
   hipStream_t st = (hipStream_t)stream;
+  int s2epi = 1;
+  {
+    const char* e = getenv("DDLW_S2_EPI");
+    if (e && e[0] == '0') s2epi = 0;
+  }
 #define S2LAUNCH(BN_)                                                          \
   do {                                                                         \
     long gx = cdiv(M, 128) * cdiv(Cdx, BN_);                                   \
-    hipLaunchKernelGGL((k_conv_fwd_igemm<128, BN_, true, 2, false>),           \
+    if (!s2epi)                                                                \
+      hipLaunchKernelGGL((k_conv_fwd_igemm<128, BN_, false, 2, false>),        \
+                         dim3((int)gx, 4), dim3(256), 0, st,                   \
+                         (const bf16_t*)dy, (const bf16_t*)wcat, (bf16_t*)dx,  \
+                         (const bf16_t*)zpage, N, Ho, Wo, Kdy, Cdx, oh, ow, 2, \
+                         2, 1, 0, (int)gx, ih, iw, 2, mg_wo, sh_wo, mg_ho,     \
+                         sh_ho, nullptr, nullptr, nullptr, nullptr, nullptr,   \
+                         nullptr, nullptr, 1, q);                              \
+    else                                                                       \
+      hipLaunchKernelGGL((k_conv_fwd_igemm<128, BN_, true, 2, false>),         \
                        dim3((int)gx, 4), dim3(256), 0, st, (const bf16_t*)dy,  \
                        (const bf16_t*)wcat, (bf16_t*)dx, (const bf16_t*)zpage, \
                        N, Ho, Wo, Kdy, Cdx, oh, ow, 2, 2, 1, 0, (int)gx, ih,   \
