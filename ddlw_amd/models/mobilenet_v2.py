@@ -123,7 +123,9 @@ class TransferModel(nn.Module):
         super().__init__()
         self.base = FrozenBase(MobileNetV2(channels))
         self.global_average_pooling = GlobalAvgPool2d()
-        self.dropout = nn.Dropout(dropout)
+        from ..ops.layers import Dropout as DdlwDropout
+
+        self.dropout = DdlwDropout(dropout)  # K7 kernel on GPU, stock on CPU
         self.classifier = nn.Linear(1280, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

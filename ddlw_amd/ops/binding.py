@@ -326,3 +326,55 @@ def depthwise_fwd(x: torch.Tensor, weight: torch.Tensor, stride: int, padding: i
         "depthwise_fwd",
     )
     return y
+
+
+def dropout_fwd(x: torch.Tensor, p: float, seed: int):
+    """K7: bitmask dropout (counter-based RNG, deterministic per seed).
+    Returns (y, mask); x flat-size must be a multiple of 8."""
+    lib = require_lib()
+    n = x.numel()
+    y = torch.empty_like(x)
+    mask = torch.empty(n // 8, dtype=torch.uint8, device=x.device)
+    check(lib.ddlw_dropout_fwd(_p(x), _p(y), _p(mask), ctypes.c_long(n),
+                               ctypes.c_float(p),
+                               ctypes.c_ulonglong(seed & (2**64 - 1)),
+                               ctypes.c_void_p(current_stream_ptr())),
+          "dropout_fwd")
+    return y, mask
+
+
+def dropout_bwd(dy: torch.Tensor, mask: torch.Tensor, p: float):
+    lib = require_lib()
+    dx = torch.empty_like(dy)
+    check(lib.ddlw_dropout_bwd(_p(dy), _p(mask), _p(dx),
+                               ctypes.c_long(dy.numel()), ctypes.c_float(p),
+                               ctypes.c_void_p(current_stream_ptr())),
+          "dropout_bwd")
+    return dx
+
+
+def argmax_rows(logits: torch.Tensor) -> torch.Tensor:
+    """K12: per-row argmax (first-max tie-break, like torch.argmax)."""
+    lib = require_lib()
+    lf = logits.float().contiguous()
+    n, c = lf.shape
+    out = torch.empty(n, dtype=torch.long, device=lf.device)
+    check(lib.ddlw_argmax_rows(_p(lf), _p(out), ctypes.c_long(n),
+                               ctypes.c_int(c),
+                               ctypes.c_void_p(current_stream_ptr())),
+          "argmax_rows")
+    return out
+
+
+def accuracy(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """K11: mean(argmax(logits) == labels) via per-block partial counts."""
+    lib = require_lib()
+    lf = logits.float().contiguous()
+    n, c = lf.shape
+    nblocks = (n + 3) // 4
+    partial = torch.empty(nblocks, dtype=torch.int32, device=lf.device)
+    check(lib.ddlw_accuracy(_p(lf), _p(labels.contiguous()), _p(partial),
+                            ctypes.c_long(n), ctypes.c_int(c),
+                            ctypes.c_void_p(current_stream_ptr())),
+          "accuracy")
+    return partial.sum().float() / n

@@ -1012,3 +1012,140 @@ DDLW_EXPORT int ddlw_depthwise_fwd(const void* x, const void* w_t, void* y,
                      (bf16_t*)y, N, H, W_, C, Ho, Wo, R, S, stride, pad);
   DDLW_CHECK_LAUNCH();
 }
+
+// ---------------------------------------------------------------------------
+// K7: Dropout fwd/bwd (head path, B x 1280 — reference P1/02:174, p tunable
+// P2/01:197). Counter-based RNG (splitmix64 of (seed, index)): stateless,
+// deterministic given the seed, no curand dependency. Bitmask layout = one
+// byte per 8 elements (same convention as the BN ReLU mask).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ unsigned ddlw_hash32(unsigned long long s) {
+  s ^= s >> 33;
+  s *= 0xff51afd7ed558ccdULL;
+  s ^= s >> 33;
+  s *= 0xc4ceb9fe1a85ec53ULL;
+  s ^= s >> 33;
+  return (unsigned)s;
+}
+
+__global__ __launch_bounds__(256) void k_dropout_fwd(
+    const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+    unsigned char* __restrict__ mask, long n8, float p, float scale,
+    unsigned long long seed) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 v, o;
+    v.v = *reinterpret_cast<const uint4*>(x + i * 8);
+    unsigned char mb = 0;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      unsigned r = ddlw_hash32(seed + (unsigned long long)(i * 8 + k));
+      bool keep = (float)(r >> 8) * (1.f / 16777216.f) >= p;
+      if (keep) mb |= (1u << k);
+      o.h[k] = keep ? f2b(b2f(v.h[k]) * scale) : (bf16_t)0;
+    }
+    *reinterpret_cast<uint4*>(y + i * 8) = o.v;
+    mask[i] = mb;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_dropout_bwd(
+    const bf16_t* __restrict__ dy, const unsigned char* __restrict__ mask,
+    bf16_t* __restrict__ dx, long n8, float scale) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 v, o;
+    v.v = *reinterpret_cast<const uint4*>(dy + i * 8);
+    unsigned char mb = mask[i];
+    #pragma unroll
+    for (int k = 0; k < 8; ++k)
+      o.h[k] = ((mb >> k) & 1) ? f2b(b2f(v.h[k]) * scale) : (bf16_t)0;
+    *reinterpret_cast<uint4*>(dx + i * 8) = o.v;
+  }
+}
+
+DDLW_EXPORT int ddlw_dropout_fwd(const void* x, void* y, void* mask, long n,
+                                 float p, unsigned long long seed,
+                                 void* stream) {
+  if (n % 8 != 0) {
+    ddlw_set_error("dropout: n must be a multiple of 8");
+    return 2;
+  }
+  float scale = 1.f / (1.f - p);
+  hipLaunchKernelGGL(k_dropout_fwd, dim3(grid_1d(n / 8)), dim3(256), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)y,
+                     (unsigned char*)mask, n / 8, p, scale, seed);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_dropout_bwd(const void* dy, const void* mask, void* dx,
+                                 long n, float p, void* stream) {
+  float scale = 1.f / (1.f - p);
+  hipLaunchKernelGGL(k_dropout_bwd, dim3(grid_1d(n / 8)), dim3(256), 0,
+                     (hipStream_t)stream, (const bf16_t*)dy,
+                     (const unsigned char*)mask, (bf16_t*)dx, n / 8, scale);
+  DDLW_CHECK_LAUNCH();
+}
+
+// ---------------------------------------------------------------------------
+// K12: row argmax (inference: logits -> class index, P2/03:208-210) and
+// K11: accuracy (argmax == label, per-block deterministic partial counts).
+// One wave per row; first-max-index tie-break matches torch.argmax.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int ddlw_row_argmax(const float* row, int C) {
+  const int lane = threadIdx.x & 63;
+  float best = -INFINITY;
+  int bi = 0x7fffffff;
+  for (int c = lane; c < C; c += 64) {
+    float v = row[c];
+    if (v > best) { best = v; bi = c; }
+  }
+  #pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    float ov = __shfl_down(best, off, 64);
+    int oi = __shfl_down(bi, off, 64);
+    if (ov > best || (ov == best && oi < bi)) { best = ov; bi = oi; }
+  }
+  return bi;  // valid in lane 0
+}
+
+__global__ __launch_bounds__(256) void k_argmax_rows(
+    const float* __restrict__ logits, long* __restrict__ out, long N, int C) {
+  long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= N) return;
+  int bi = ddlw_row_argmax(logits + row * C, C);
+  if ((threadIdx.x & 63) == 0) out[row] = bi;
+}
+
+__global__ __launch_bounds__(256) void k_accuracy(
+    const float* __restrict__ logits, const long* __restrict__ labels,
+    int* __restrict__ partial, long N, int C) {
+  long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  __shared__ int cnt[4];
+  if (threadIdx.x < 4) cnt[threadIdx.x] = 0;
+  __syncthreads();
+  if (row < N) {
+    int bi = ddlw_row_argmax(logits + row * C, C);
+    if ((threadIdx.x & 63) == 0)
+      cnt[threadIdx.x >> 6] = (bi == (int)labels[row]) ? 1 : 0;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0)
+    partial[blockIdx.x] = cnt[0] + cnt[1] + cnt[2] + cnt[3];
+}
+
+DDLW_EXPORT int ddlw_argmax_rows(const void* logits, void* out, long N, int C,
+                                 void* stream) {
+  hipLaunchKernelGGL(k_argmax_rows, dim3((int)((N + 3) / 4)), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)logits, (long*)out, N,
+                     C);
+  DDLW_CHECK_LAUNCH();
+}
+
+DDLW_EXPORT int ddlw_accuracy(const void* logits, const void* labels,
+                              void* partial, long N, int C, void* stream) {
+  hipLaunchKernelGGL(k_accuracy, dim3((int)((N + 3) / 4)), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)logits,
+                     (const long*)labels, (int*)partial, N, C);
+  DDLW_CHECK_LAUNCH();
+}

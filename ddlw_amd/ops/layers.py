@@ -150,6 +150,49 @@ class BatchNormAct2d(nn.Module):
 
 
 # --------------------------------------------------------------------------- #
+# Dropout (K7 — head path, reference P1/02:174)
+# --------------------------------------------------------------------------- #
+
+
+class _DropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p: float, seed: int):
+        y, mask = binding.dropout_fwd(x.contiguous(), p, seed)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        return binding.dropout_bwd(dy.contiguous(), mask, ctx.p), None, None
+
+
+class Dropout(nn.Module):
+    """Bitmask dropout on the ddlw kernel (counter-based RNG, deterministic
+    per (base_seed, call_counter)); stock fallback off-GPU / odd sizes."""
+
+    def __init__(self, p: float = 0.5):
+        super().__init__()
+        self.p = float(p)
+        self._calls = 0
+        self._base_seed = int(torch.initial_seed()) & (2**63 - 1)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if not self.training or self.p <= 0.0:
+            return x
+        if (x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 8 == 0
+                and not _hip_ops_disabled()):
+            self._calls += 1
+            seed = self._base_seed + self._calls * 0x9E3779B97F4A7C15
+            return _DropoutFn.apply(x, self.p, seed)
+        return F.dropout(x, self.p, self.training)
+
+    def extra_repr(self) -> str:
+        return f"p={self.p}"
+
+
+# --------------------------------------------------------------------------- #
 # MaxPool 3x3 s2 p1
 # --------------------------------------------------------------------------- #
 

@@ -81,6 +81,18 @@ class History:
             self.history.setdefault(k, []).append(v)
 
 
+def _accuracy(logits, labels):
+    """K11 on the hip kernel for CUDA logits; torch ops otherwise."""
+    if logits.is_cuda and logits.dim() == 2:
+        try:
+            from ..ops import binding
+
+            return binding.accuracy(logits, labels)
+        except Exception:
+            pass
+    return (logits.argmax(-1) == labels).float().mean()
+
+
 class Model:
     """Wraps an ``nn.Module`` with compile/fit/evaluate."""
 
@@ -131,7 +143,7 @@ class Model:
         self.optimizer.step()
         out = {"loss": float(loss.detach())}
         if "accuracy" in self.metrics:
-            out["accuracy"] = float((logits.detach().argmax(-1) == labels).float().mean())
+            out["accuracy"] = float(_accuracy(logits.detach(), labels))
         return out
 
     @torch.no_grad()
@@ -141,7 +153,7 @@ class Model:
         logits, loss = self._forward_loss(images, labels)
         out = {"loss": float(loss)}
         if "accuracy" in self.metrics:
-            out["accuracy"] = float((logits.argmax(-1) == labels).float().mean())
+            out["accuracy"] = float(_accuracy(logits, labels))
         return out
 
     # ------------------------------------------------------------------ #

@@ -353,3 +353,56 @@ def test_training_equivalence_hip_vs_stock():
     for a, b in zip(hip, ref):
         assert abs(a - b) / max(abs(b), 0.3) < 0.35, (hip, ref)
 
+
+
+def test_dropout_kernel_stats_and_determinism():
+    """K7: keep-rate ~ (1-p), kept values scaled 1/(1-p), deterministic per
+    seed, and backward zeroes exactly the dropped positions."""
+    from ddlw_amd.ops import binding
+
+    torch.manual_seed(11)
+    x = torch.randn(64, 1280, device="cuda").to(torch.bfloat16).abs() + 0.5
+    p = 0.37
+    y1, m1 = binding.dropout_fwd(x, p, seed=1234)
+    y2, m2 = binding.dropout_fwd(x, p, seed=1234)
+    assert torch.equal(y1, y2) and torch.equal(m1, m2)  # deterministic
+    y3, _ = binding.dropout_fwd(x, p, seed=99)
+    assert not torch.equal(y1, y3)  # seed-dependent
+    kept = (y1 != 0)
+    rate = kept.float().mean().item()
+    assert abs(rate - (1 - p)) < 0.02, rate
+    scale = (y1.float()[kept] / x.float()[kept])
+    assert ((scale - 1 / (1 - p)).abs() < 0.02).all()
+    dy = torch.ones_like(x)
+    dx = binding.dropout_bwd(dy, m1, p)
+    assert torch.equal((dx != 0), kept)  # same mask both ways
+
+
+def test_dropout_module_autograd():
+    from ddlw_amd.ops.layers import Dropout
+
+    torch.manual_seed(3)
+    d = Dropout(0.5).train()
+    x = torch.randn(8, 1280, device="cuda").to(torch.bfloat16).requires_grad_(True)
+    y = d(x)
+    y.float().sum().backward()
+    # grad nonzero exactly where output nonzero
+    assert torch.equal((x.grad != 0), (y != 0))
+    d.eval()
+    assert torch.equal(d(x), x)  # identity in eval
+
+
+def test_argmax_and_accuracy_kernels():
+    """K11/K12 vs torch (incl. first-max tie-break)."""
+    from ddlw_amd.ops import binding
+
+    torch.manual_seed(5)
+    for C in (5, 33, 1000):
+        logits = torch.randn(257, C, device="cuda")
+        logits[0, :] = 1.0  # all-ties row: torch picks index 0
+        idx = binding.argmax_rows(logits)
+        assert torch.equal(idx, logits.argmax(-1))
+        labels = torch.randint(0, C, (257,), device="cuda")
+        acc = binding.accuracy(logits, labels)
+        ref = (logits.argmax(-1) == labels).float().mean()
+        assert torch.allclose(acc, ref), (C, acc, ref)
