@@ -18,6 +18,15 @@ class ChromeTracer:
         self.pid = pid if pid is not None else os.getpid()
         self._events = []
         self._lock = threading.Lock()
+        if self.enabled:
+            # one file per rank (multi-process DP), dumped at exit — like
+            # HOROVOD_TIMELINE, no explicit save needed from user code
+            rank = os.environ.get("RANK")
+            if rank is not None:
+                self.path = f"{self.path}.rank{rank}"
+            import atexit
+
+            atexit.register(self.save)
 
     def event(self, name: str, cat: str, t0_us: float, dur_us: float, tid: int = 0, args: Optional[dict] = None):
         if not self.enabled:
