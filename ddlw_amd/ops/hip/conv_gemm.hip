@@ -414,12 +414,14 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           int c8 = (kh * 4 + fr_c8) ^ ((row >> 1) & 7);
           fb[ni] = *reinterpret_cast<const bf16x8_v*>(lB + row * BK + c8 * 8);
         }
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int mi = 0; mi < MF; ++mi)
           #pragma unroll
           for (int ni = 0; ni < NF; ++ni)
             acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 fa[mi], fb[ni], acc[mi][ni], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
       }
     }
     __syncthreads();
