@@ -16,6 +16,8 @@ from __future__ import annotations
 
 from typing import List
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -46,6 +48,9 @@ class Bottleneck(nn.Module):
             # whole-block fused Function: the residual-join gradient add is
             # absorbed into conv1's dgrad epilogue (see ops/block.py)
             return _block.bottleneck_forward(self, x)
+        if _block.bottleneck_eval_fusable(self, x):
+            # eval: BN folded into the conv epilogues (4 kernels per block)
+            return _block.bottleneck_eval_forward(self, x)
         identity = self.downsample(x) if self.downsample is not None else x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
@@ -102,7 +107,23 @@ class ResNet50(nn.Module):
                 nn.init.zeros_(m.bn3.weight)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.bn1(self.conv1(x))
+        from ..ops import block as _block
+
+        if (not self.training and not torch.is_grad_enabled() and x.is_cuda
+                and x.dtype == torch.bfloat16
+                and os.environ.get("DDLW_EVAL_FOLD", "1") == "1"
+                and os.environ.get("DDLW_STEM", "1") == "1"
+                and os.environ.get("DDLW_DISABLE_HIP_OPS", "0") != "1"
+                and self.conv1.in_channels == 3):
+            from ..ops import conv_gemm as _cg
+
+            # eval: stem conv + folded bn1 + relu in one kernel
+            x = _cg.stem_fwd_kernel(
+                x.contiguous(memory_format=torch.channels_last),
+                self.conv1.weight,
+                ep=(*self.bn1.folded_scale_bias(), True))
+        else:
+            x = self.bn1(self.conv1(x))
         x = self.maxpool(x)
         x = self.layer1(x)
         x = self.layer2(x)

@@ -239,6 +239,60 @@ class _BottleneckFn(torch.autograd.Function):
                 dwd, dgd, dbd, None)
 
 
+@torch.no_grad()
+def bottleneck_eval_forward(block, x: torch.Tensor) -> torch.Tensor:
+    """Eval-mode bottleneck: BN folded into the conv epilogues — 4 kernels
+    per block (vs conv+bn_apply pairs). Running stats are constants in
+    eval, so y = relu(conv*scale + bias [+ res]) is exact BN semantics."""
+    x = _cl(x)
+    mode = os.environ.get("DDLW_CONV", "auto")
+    stride = block.stride
+
+    def _conv_ep(inp, conv, w, bn, relu, acc=None, st=1, pad=0):
+        conv_gemm.available(conv, inp, mode)
+        rf = getattr(conv, "_ddlw_route", (False, False, False))[0]
+        if rf:
+            return conv_gemm.conv_fwd_kernel(
+                inp, w.to(torch.bfloat16), st, pad, acc=acc,
+                ep=(*bn.folded_scale_bias(), relu))
+        t = torch.nn.functional.conv2d(inp, w.to(inp.dtype), None, st, pad)
+        s, b = bn.folded_scale_bias()
+        y = t.float() * s.view(1, -1, 1, 1) + b.view(1, -1, 1, 1)
+        if acc is not None:
+            y = y + acc.float()
+        if relu:
+            y = torch.relu(y)
+        return _cl(y.to(inp.dtype))
+
+    if block.downsample is not None:
+        ds = block.downsample
+        res = _conv_ep(x, ds.conv, ds.conv.weight, ds.bn, False,
+                       st=stride, pad=0)
+    else:
+        res = x
+    a1 = _conv_ep(x, block.conv1, block.conv1.weight, block.bn1, True)
+    a2 = _conv_ep(a1, block.conv2, block.conv2.weight, block.bn2, True,
+                  st=stride, pad=1)
+    return _conv_ep(a2, block.conv3, block.conv3.weight, block.bn3, True,
+                    acc=res)
+
+
+def bottleneck_eval_fusable(block, x: torch.Tensor) -> bool:
+    if block.training or torch.is_grad_enabled():
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    if os.environ.get("DDLW_DISABLE_HIP_OPS", "0") == "1":
+        return False
+    if os.environ.get("DDLW_EVAL_FOLD", "1") != "1":
+        return False
+    for bn in [block.bn1, block.bn2, block.bn3] + (
+            [block.downsample.bn] if block.downsample is not None else []):
+        if bn.running_mean.dtype != torch.float32:
+            return False
+    return True
+
+
 def bottleneck_fusable(block, x: torch.Tensor) -> bool:
     """Fused path: training, grad-enabled, bf16 CUDA input, fp32 BN
     params/stats, kernel-supported channel counts."""

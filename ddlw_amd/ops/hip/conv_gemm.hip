@@ -208,6 +208,9 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     const unsigned char* __restrict__ bnb_mask,  // become (dbeta, dgamma)
     const float* __restrict__ bnb_mean,
     const float* __restrict__ bnb_rstd,
+    const float* __restrict__ ep_scale = nullptr,  // eval-BN fold:
+    const float* __restrict__ ep_bias = nullptr,   //  y = relu?(v*s+b [+acc])
+    int ep_relu = 0,
     int s2_merged = 0, S2Quad quad = {}) {
   if (s2_merged) {
     const S2Class& cc = quad.c[blockIdx.y];
@@ -476,6 +479,15 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
         }
       }
     } else {
+      float eps_[NF], epb_[NF];
+      if (ep_scale) {
+        #pragma unroll
+        for (int ni = 0; ni < NF; ++ni) {
+          int j = tile_n * BN + wc * WN + ni * 16 + d_col;
+          eps_[ni] = (j < K) ? ep_scale[j] : 0.f;
+          epb_[ni] = (j < K) ? ep_bias[j] : 0.f;
+        }
+      }
       #pragma unroll
       for (int mi = 0; mi < MF; ++mi) {
         #pragma unroll
@@ -488,7 +500,9 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
             if (m < M) {
               long oi = out_row(m) * K + j;
               float v = acc[mi][ni][q];
+              if (ep_scale) v = v * eps_[ni] + epb_[ni];
               if (accp) v += b2f(accp[oi]);
+              if (ep_relu) v = fmaxf(v, 0.f);
               y[oi] = f2b_hw(v);
             }
           }
@@ -536,6 +550,21 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
   const int e_ch = lane % CPL;       // chunk within the row
   const long m_base = tile_m * BM + wr * WM;
   const int j_base = tile_n * BN + wc * WN + e_ch * 8;
+  float eps8[8], epb8[8];
+  if (ep_scale) {
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int j = j_base + e;
+      eps8[e] = (j < K) ? ep_scale[j] : 0.f;
+      epb8[e] = (j < K) ? ep_bias[j] : 0.f;
+    }
+  }
+  // apply the eval-BN fold (+acc +relu) to one bf16 element
+  auto ep_apply = [&](bf16_t v, int e, float accv) -> bf16_t {
+    float f = b2f(v) * eps8[e] + epb8[e] + accv;
+    if (ep_relu) f = fmaxf(f, 0.f);
+    return f2b_hw(f);
+  };
   float s8[8], q8[8], bnm8[8], bnr8[8];
   #pragma unroll
   for (int e = 0; e < 8; ++e) { s8[e] = 0.f; q8[e] = 0.f; }
@@ -555,7 +584,24 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
     if (m < M) {
       const long orow = out_row(m);
       if (j_base + 8 <= K) {
-        if (accp) {
+        if (ep_scale) {
+          uint4 a{0, 0, 0, 0};
+          if (accp)
+            a = *reinterpret_cast<const uint4*>(accp + orow * K + j_base);
+          uint4 o;
+          unsigned* ow = &o.x;
+          const unsigned* vw = &val.x;
+          const unsigned* aw = &a.x;
+          #pragma unroll
+          for (int d = 0; d < 4; ++d) {
+            float alo = accp ? b2f((bf16_t)(aw[d] & 0xffff)) : 0.f;
+            float ahi = accp ? b2f((bf16_t)(aw[d] >> 16)) : 0.f;
+            bf16_t lo = ep_apply((bf16_t)(vw[d] & 0xffff), 2 * d, alo);
+            bf16_t hi = ep_apply((bf16_t)(vw[d] >> 16), 2 * d + 1, ahi);
+            ow[d] = (unsigned)lo | ((unsigned)hi << 16);
+          }
+          *reinterpret_cast<uint4*>(y + orow * K + j_base) = o;
+        } else if (accp) {
           // accumulate: coalesced uint4 read of the add-input at the same
           // address, pairwise bf16 add in fp32 (matches the engine's
           // bf16+bf16 add numerics)
@@ -569,8 +615,10 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           val.y = addpair(val.y, a.y);
           val.z = addpair(val.z, a.z);
           val.w = addpair(val.w, a.w);
+          *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
+        } else {
+          *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
         }
-        *reinterpret_cast<uint4*>(y + orow * K + j_base) = val;
         if (bn_ps) {
           if (bnb_x) {
             // BN-BWD partials: masked dy and dy*xhat (j_base is 8-aligned
@@ -607,7 +655,10 @@ __global__ __launch_bounds__(256) void k_conv_fwd_igemm(
           if (j_base + e < K) {
             unsigned wd = (e < 2) ? val.x : (e < 4) ? val.y : (e < 6) ? val.z : val.w;
             bf16_t ov = (bf16_t)(wd >> ((e & 1) * 16));
-            if (accp)
+            if (ep_scale) {
+              float av = accp ? b2f(accp[orow * K + j_base + e]) : 0.f;
+              ov = ep_apply(ov, e, av);
+            } else if (accp)
               ov = f2b_hw(b2f(ov) + b2f(accp[orow * K + j_base + e]));
             y[orow * K + j_base + e] = ov;
             if (bn_ps) {
@@ -935,7 +986,7 @@ struct TilePick {
 // construction). want_stats forces the 16x16 direct epilogue (the only
 // ones carrying the bn-partials code).
 static TilePick pick_tile(long M, int K, int C, int T, bool want_stats,
-                          bool want_bnb = false) {
+                          bool want_bnb = false, bool want_ep = false) {
   int mfma_pref, wide_pref;
   {
     const char* e = getenv("DDLW_CONV_MFMA");
@@ -953,7 +1004,7 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats,
   // isolated win). Auto routing is therefore OFF — a documented negative
   // result; DDLW_CONV_WIDE=1 forces it for probes.
   const bool wide_auto = false;
-  if (!want_bnb &&
+  if (!want_bnb && !want_stats && !want_ep &&
       ((wide_pref == 1 && K >= 128 && T >= 4) || (wide_pref == -1 && wide_auto))) {
     p.wide = true;
     p.bm = 256;
@@ -986,6 +1037,8 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
                            void* bn_ps, void* bn_pq,
                            const void* bnb_x, const void* bnb_mask,
                            const void* bnb_mean, const void* bnb_rstd,
+                           const void* ep_scale, const void* ep_bias,
+                           int ep_relu,
                            int N, int H, int W_, int C, int K,
                            int Ho, int Wo, int R, int S, int stride,
                            int pad, int oH, int oW, int oS, void* stream) {
@@ -1006,7 +1059,9 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
   hipStream_t st = (hipStream_t)stream;
   const bool want_stats = bn_ps != nullptr;
   const bool want_bnb = bnb_x != nullptr;
-  TilePick p = pick_tile(M, K, C, T, want_stats, want_bnb);
+  // the eval-BN fold epilogue lives only in the base kernel
+  const bool want_ep = ep_scale != nullptr;
+  TilePick p = pick_tile(M, K, C, T, want_stats, want_bnb, want_ep);
   if (p.wide) {
     long grid = cdiv(M, 256) * cdiv(K, p.bn);
 #define WIDE_LAUNCH(BN_, NB_)                                                  \
@@ -1037,7 +1092,9 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
                          (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq,    \
                          (const bf16_t*)bnb_x,                                \
                          (const unsigned char*)bnb_mask,                      \
-                         (const float*)bnb_mean, (const float*)bnb_rstd);     \
+                         (const float*)bnb_mean, (const float*)bnb_rstd,      \
+                         (const float*)ep_scale, (const float*)ep_bias,       \
+                         ep_relu);                                            \
     else                                                                      \
       hipLaunchKernelGGL((k_conv_fwd_igemm<BM, BN, EPI, BUFS, true>),         \
                          dim3((int)grid), dim3(256), 0, st, (const bf16_t*)x, \
@@ -1047,7 +1104,9 @@ static int conv_fwd_launch(const void* x, const void* w, void* y,
                          (const bf16_t*)acc, (float*)bn_ps, (float*)bn_pq,    \
                          (const bf16_t*)bnb_x,                                \
                          (const unsigned char*)bnb_mask,                      \
-                         (const float*)bnb_mean, (const float*)bnb_rstd);     \
+                         (const float*)bnb_mean, (const float*)bnb_rstd,      \
+                         (const float*)ep_scale, (const float*)ep_bias,       \
+                         ep_relu);                                            \
   } while (0)
   if (p.bn == 128) {
     if (p.epi_lds) {
@@ -1085,8 +1144,9 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_acc(const void* x, const void* w, void* y,
                                         int pad, int oH, int oW, int oS,
                                         void* stream) {
   return conv_fwd_launch(x, w, y, zpage, acc, nullptr, nullptr, nullptr,
-                         nullptr, nullptr, nullptr, N, H, W_, C, K, Ho, Wo, R,
-                         S, stride, pad, oH, oW, oS, stream);
+                         nullptr, nullptr, nullptr, nullptr, nullptr, 0, N, H,
+                         W_, C, K, Ho, Wo, R, S, stride, pad, oH, oW, oS,
+                         stream);
 }
 
 DDLW_EXPORT int ddlw_conv_fwd_igemm_stats(const void* x, const void* w, void* y,
@@ -1096,8 +1156,9 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_stats(const void* x, const void* w, void* y,
                                           int Ho, int Wo, int R, int S,
                                           int stride, int pad, void* stream) {
   return conv_fwd_launch(x, w, y, zpage, nullptr, bn_ps, bn_pq, nullptr,
-                         nullptr, nullptr, nullptr, N, H, W_, C, K, Ho, Wo, R,
-                         S, stride, pad, Ho, Wo, 1, stream);
+                         nullptr, nullptr, nullptr, nullptr, nullptr, 0, N, H,
+                         W_, C, K, Ho, Wo, R, S, stride, pad, Ho, Wo, 1,
+                         stream);
 }
 
 // dgrad + fused BN-backward reduce: y = dgrad output (the BN's dy); the
@@ -1110,8 +1171,25 @@ DDLW_EXPORT int ddlw_conv_fwd_igemm_bnb(
     int N, int H, int W_, int C, int K, int Ho, int Wo, int R, int S,
     int stride, int pad, void* stream) {
   return conv_fwd_launch(x, w, y, zpage, acc, p_db, p_dg, bnb_x, bnb_mask,
-                         bnb_mean, bnb_rstd, N, H, W_, C, K, Ho, Wo, R, S,
-                         stride, pad, Ho, Wo, 1, stream);
+                         bnb_mean, bnb_rstd, nullptr, nullptr, 0, N, H, W_, C,
+                         K, Ho, Wo, R, S, stride, pad, Ho, Wo, 1, stream);
+}
+
+// fwd + fused eval-mode BN apply: y = relu?(conv*scale[k] + bias[k] [+ acc])
+// — scale/bias are the host-folded running-stat BN transform, so the whole
+// bn_apply pass disappears from inference/eval (and the acc input carries
+// the residual shortcut for the bn3-join).
+DDLW_EXPORT int ddlw_conv_fwd_igemm_ep(const void* x, const void* w, void* y,
+                                       const void* zpage, const void* acc,
+                                       const void* ep_scale, const void* ep_bias,
+                                       int ep_relu,
+                                       int N, int H, int W_, int C, int K,
+                                       int Ho, int Wo, int R, int S,
+                                       int stride, int pad, void* stream) {
+  return conv_fwd_launch(x, w, y, zpage, acc, nullptr, nullptr, nullptr,
+                         nullptr, nullptr, nullptr, ep_scale, ep_bias, ep_relu,
+                         N, H, W_, C, K, Ho, Wo, R, S, stride, pad, Ho, Wo, 1,
+                         stream);
 }
 
 // merged-class stride-2 3x3 dgrad (even ih/iw): dx = scattered union of 4
@@ -1154,14 +1232,15 @@ DDLW_EXPORT int ddlw_conv_dgrad_s2m(const void* dy, const void* wcat, void* dx,
                          (const bf16_t*)zpage, N, Ho, Wo, Kdy, Cdx, oh, ow, 2, \
                          2, 1, 0, (int)gx, ih, iw, 2, mg_wo, sh_wo, mg_ho,     \
                          sh_ho, nullptr, nullptr, nullptr, nullptr, nullptr,   \
-                         nullptr, nullptr, 1, q);                              \
+                         nullptr, nullptr, nullptr, nullptr, 0, 1, q);         \
     else                                                                       \
       hipLaunchKernelGGL((k_conv_fwd_igemm<128, BN_, true, 2, false>),         \
                        dim3((int)gx, 4), dim3(256), 0, st, (const bf16_t*)dy,  \
                        (const bf16_t*)wcat, (bf16_t*)dx, (const bf16_t*)zpage, \
                        N, Ho, Wo, Kdy, Cdx, oh, ow, 2, 2, 1, 0, (int)gx, ih,   \
                        iw, 2, mg_wo, sh_wo, mg_ho, sh_ho, nullptr, nullptr,    \
-                       nullptr, nullptr, nullptr, nullptr, nullptr, 1, q);     \
+                       nullptr, nullptr, nullptr, nullptr, nullptr, nullptr,   \
+                       nullptr, 0, 1, q);     \
   } while (0)
   if (Cdx >= 128)
     S2LAUNCH(128);

@@ -56,7 +56,9 @@ __global__ __launch_bounds__(256) void k_conv_stem(
     const bf16_t* __restrict__ zpage,
     int N, int H, int Wp, int K, int Ho, int Wo, int stride, int pad,
     int nwg_swz, unsigned long long magic_wo, unsigned shift_wo,
-    unsigned long long magic_ho, unsigned shift_ho) {
+    unsigned long long magic_ho, unsigned shift_ho,
+    const float* __restrict__ ep_scale, const float* __restrict__ ep_bias,
+    int ep_relu) {
   constexpr int BM = 128, BN = 64, BK = 64, T = 4;
   constexpr int WM = 64, WN = 32, MF = WM / 16, NF = WN / 16;
   constexpr int AP = BM / 32;  // 1-KiB A pieces per wave (8 rows each)
@@ -202,13 +204,33 @@ __global__ __launch_bounds__(256) void k_conv_stem(
   const int e_ch = lane % CPL;
   const long m_base = tile_m * BM + wr * WM;
   const int j_base = wc * WN + e_ch * 8;
+  float eps8[8], epb8[8];
+  if (ep_scale) {
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      eps8[e] = ep_scale[j_base + e];
+      epb8[e] = ep_bias[j_base + e];
+    }
+  }
   #pragma unroll
   for (int it = 0; it < WM / RPI; ++it) {
     const int row = it * RPI + e_row;
     const long m = m_base + row;
     uint4 val = *reinterpret_cast<const uint4*>(lC + row * WNP + e_ch * 8);
-    if (m < M)
+    if (m < M) {
+      if (ep_scale) {
+        // fused eval-BN apply (+relu): the stem's bn1 disappears
+        unsigned* vw = &val.x;
+        #pragma unroll
+        for (int d = 0; d < 4; ++d) {
+          float lo = b2f((bf16_t)(vw[d] & 0xffff)) * eps8[2 * d] + epb8[2 * d];
+          float hi = b2f((bf16_t)(vw[d] >> 16)) * eps8[2 * d + 1] + epb8[2 * d + 1];
+          if (ep_relu) { lo = fmaxf(lo, 0.f); hi = fmaxf(hi, 0.f); }
+          vw[d] = (unsigned)f2b(lo) | ((unsigned)f2b(hi) << 16);
+        }
+      }
       *reinterpret_cast<uint4*>(y + m * K + j_base) = val;
+    }
   }
 }
 
@@ -238,7 +260,8 @@ DDLW_EXPORT int ddlw_stem_repack(const void* x, void* x4, int N, int H, int W,
 DDLW_EXPORT int ddlw_conv_stem(const void* x4, const void* w4, void* y,
                                const void* zpage, int N, int H, int W, int K,
                                int Ho, int Wo, int stride, int pad,
-                               void* stream) {
+                               const void* ep_scale, const void* ep_bias,
+                               int ep_relu, void* stream) {
   if (K != 64 || stride != 2 || pad != 3) {
     ddlw_set_error("conv_stem: supports K=64, stride=2, pad=3 (7x7 stem)");
     return 2;
@@ -254,6 +277,7 @@ DDLW_EXPORT int ddlw_conv_stem(const void* x4, const void* w4, void* y,
                      (hipStream_t)stream, (const bf16_t*)x4,
                      (const bf16_t*)w4, (bf16_t*)y, (const bf16_t*)zpage, N,
                      H, Wp, K, Ho, Wo, stride, pad, (int)grid, mg_wo, sh_wo,
-                     mg_ho, sh_ho);
+                     mg_ho, sh_ho, (const float*)ep_scale,
+                     (const float*)ep_bias, ep_relu);
   DDLW_CHECK_LAUNCH();
 }

@@ -145,6 +145,23 @@ class BatchNormAct2d(nn.Module):
             y = y + residual
         return F.relu(y) if self.relu else y
 
+    def folded_scale_bias(self):
+        """Eval-mode BN as one affine transform: y = x*scale + bias with
+        scale = gamma*rsqrt(rv+eps), bias = beta - rm*scale. Cached until
+        any source tensor mutates (torch _version counters)."""
+        key = (self.weight._version, self.bias._version,
+               self.running_mean._version, self.running_var._version)
+        cached = getattr(self, "_fold_cache", None)
+        if cached is not None and cached[0] == key:
+            return cached[1], cached[2]
+        with torch.no_grad():
+            scale = (self.weight.float()
+                     * (self.running_var.float() + self.eps).rsqrt())
+            bias = self.bias.float() - self.running_mean.float() * scale
+        scale, bias = scale.contiguous(), bias.contiguous()
+        self._fold_cache = (key, scale, bias)
+        return scale, bias
+
     def extra_repr(self) -> str:
         return f"{self.num_features}, relu={self.relu}"
 

@@ -366,3 +366,38 @@ def test_conv_dgrad_fused_bnb_parity():
         sg = dg_u.abs().max() + 1e-3
         assert ((db_f - db_u).abs().max() / sb).item() < 1e-3, (R, pad)
         assert ((dg_f - dg_u).abs().max() / sg).item() < 1e-3, (R, pad)
+
+
+def test_eval_bn_fold_parity():
+    """Eval-mode bottleneck + stem with BN folded into the conv epilogues
+    vs the unfused eval path (bn_apply kernels)."""
+    import copy
+
+    from ddlw_amd.models import build_resnet50
+
+    torch.manual_seed(12)
+    m = build_resnet50(num_classes=10).to(_cuda()).to(memory_format=torch.channels_last)
+    # non-trivial BN stats so the fold matters
+    for mod in m.modules():
+        if hasattr(mod, "running_mean") and isinstance(getattr(mod, "running_mean", None), torch.Tensor):
+            torch.nn.init.uniform_(mod.running_mean, -0.3, 0.3)
+            torch.nn.init.uniform_(mod.running_var, 0.5, 1.5)
+            torch.nn.init.uniform_(mod.weight, 0.5, 1.5)
+            torch.nn.init.uniform_(mod.bias, -0.2, 0.2)
+    for mod in m.modules():
+        if isinstance(mod, torch.nn.Conv2d):
+            mod.to(torch.bfloat16)
+    m.eval()
+    x = _cl(torch.randn(4, 3, 64, 64, device=_cuda()).to(torch.bfloat16))
+    import os as _os
+
+    with torch.no_grad():
+        y_fold = m(x)
+        _os.environ["DDLW_EVAL_FOLD"] = "0"
+        try:
+            y_ref = m(x)
+        finally:
+            _os.environ.pop("DDLW_EVAL_FOLD", None)
+    s = y_ref.float().abs().max() + 1e-6
+    rel = ((y_fold.float() - y_ref.float()).abs().max() / s).item()
+    assert rel < 5e-2, rel
