@@ -245,36 +245,21 @@ def bottleneck_eval_forward(block, x: torch.Tensor) -> torch.Tensor:
     per block (vs conv+bn_apply pairs). Running stats are constants in
     eval, so y = relu(conv*scale + bias [+ res]) is exact BN semantics."""
     x = _cl(x)
-    mode = os.environ.get("DDLW_CONV", "auto")
     stride = block.stride
 
-    def _conv_ep(inp, conv, w, bn, relu, acc=None, st=1, pad=0):
-        conv_gemm.available(conv, inp, mode)
-        rf = getattr(conv, "_ddlw_route", (False, False, False))[0]
-        if rf:
-            return conv_gemm.conv_fwd_kernel(
-                inp, w.to(torch.bfloat16), st, pad, acc=acc,
-                ep=(*bn.folded_scale_bias(), relu))
-        t = torch.nn.functional.conv2d(inp, w.to(inp.dtype), None, st, pad)
-        s, b = bn.folded_scale_bias()
-        y = t.float() * s.view(1, -1, 1, 1) + b.view(1, -1, 1, 1)
-        if acc is not None:
-            y = y + acc.float()
-        if relu:
-            y = torch.relu(y)
-        return _cl(y.to(inp.dtype))
+    def _conv_ep(inp, w, bn, relu, acc=None, st=1, pad=0):
+        return conv_gemm.conv_fwd_kernel(
+            inp, w.to(torch.bfloat16), st, pad, acc=acc,
+            ep=(*bn.folded_scale_bias(), relu))
 
     if block.downsample is not None:
         ds = block.downsample
-        res = _conv_ep(x, ds.conv, ds.conv.weight, ds.bn, False,
-                       st=stride, pad=0)
+        res = _conv_ep(x, ds.conv.weight, ds.bn, False, st=stride, pad=0)
     else:
         res = x
-    a1 = _conv_ep(x, block.conv1, block.conv1.weight, block.bn1, True)
-    a2 = _conv_ep(a1, block.conv2, block.conv2.weight, block.bn2, True,
-                  st=stride, pad=1)
-    return _conv_ep(a2, block.conv3, block.conv3.weight, block.bn3, True,
-                    acc=res)
+    a1 = _conv_ep(x, block.conv1.weight, block.bn1, True)
+    a2 = _conv_ep(a1, block.conv2.weight, block.bn2, True, st=stride, pad=1)
+    return _conv_ep(a2, block.conv3.weight, block.bn3, True, acc=res)
 
 
 def bottleneck_eval_fusable(block, x: torch.Tensor) -> bool:
@@ -286,6 +271,15 @@ def bottleneck_eval_fusable(block, x: torch.Tensor) -> bool:
         return False
     if os.environ.get("DDLW_EVAL_FOLD", "1") != "1":
         return False
+    if os.environ.get("DDLW_CONV", "auto") == "stock":
+        return False
+    # every conv must be hip-SUPPORTED (C % 64; the fwd kernel won all 22
+    # measured routes, so support — not the batch-keyed table — gates eval)
+    for conv in [block.conv1, block.conv2, block.conv3] + (
+            [block.downsample.conv] if block.downsample is not None else []):
+        if not conv_gemm.fwd_supported(conv.in_channels, conv.out_channels,
+                                       *conv.kernel_size):
+            return False
     for bn in [block.bn1, block.bn2, block.bn3] + (
             [block.downsample.bn] if block.downsample is not None else []):
         if bn.running_mean.dtype != torch.float32:

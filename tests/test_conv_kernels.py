@@ -385,7 +385,7 @@ def test_eval_bn_fold_parity():
             torch.nn.init.uniform_(mod.weight, 0.5, 1.5)
             torch.nn.init.uniform_(mod.bias, -0.2, 0.2)
     for mod in m.modules():
-        if isinstance(mod, torch.nn.Conv2d):
+        if isinstance(mod, (torch.nn.Conv2d, torch.nn.Linear)):
             mod.to(torch.bfloat16)
     m.eval()
     x = _cl(torch.randn(4, 3, 64, 64, device=_cuda()).to(torch.bfloat16))
