@@ -35,9 +35,12 @@ class ResNetPyFunc(PythonModel):
         self.model = load_torch(context.artifacts["model"])
         self.model.eval()
         if torch.cuda.is_available():
-            # fp32 params (BN stats must stay fp32); convs/fc run bf16 under
-            # autocast, activations are bf16 channels_last
             self.model = self.model.cuda().to(memory_format=torch.channels_last)
+            # bf16 conv/fc weights ONCE (BN stats stay fp32 — the eval-BN
+            # fold reads them); avoids per-batch autocast weight casts
+            for m in self.model.modules():
+                if isinstance(m, (torch.nn.Conv2d, torch.nn.Linear)):
+                    m.to(torch.bfloat16)
 
     def predict(self, context, model_input):
         import functools
@@ -73,8 +76,7 @@ class ResNetPyFunc(PythonModel):
                         pad = d[-1:].expand(bs - n_real, *d.shape[1:])
                         d = torch.cat([d, pad])  # a fresh MIOpen find
                     x = normalize_u8_bf16(d.permute(0, 3, 1, 2))
-                    with torch.autocast("cuda", dtype=torch.bfloat16):
-                        logits = self.model(x)
+                    logits = self.model(x)
                 else:
                     x = u8.permute(0, 3, 1, 2).float() / 127.5 - 1.0
                     logits = self.model(x)
