@@ -281,3 +281,214 @@ DDLW_EXPORT int ddlw_conv_stem(const void* x4, const void* w4, void* y,
                      (const float*)ep_bias, ep_relu);
   DDLW_CHECK_LAUNCH();
 }
+
+
+// ---------------------------------------------------------------------------
+// Stem weight gradient: dW4[64][256] = dy^T @ im2col(x4), reduction over
+// m = N*Ho*Wo split across blockIdx.z (same split-K + deterministic fp32
+// slab + k_wgrad_reduce pattern as conv_wgrad.hip). The 8x8x4 zero-padded
+// window is a FLAT 256-wide C dimension, so one 64x256 output tile covers
+// the whole gradient and the (r, s, c) fetch reuses the fwd repack image
+// (halo -> no horizontal bounds checks). Python slices the [:7][:7][:3]
+// real taps out of the padded result.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(2))) unsigned s_tr64_t;
+
+__global__ void k_wgrad_reduce(const float*, bf16_t*, long, int);
+
+__global__ __launch_bounds__(256) void k_stem_wgrad(
+    const bf16_t* __restrict__ dy,   // [M][64]
+    const bf16_t* __restrict__ x4,   // [N][H][Wp][4]
+    const bf16_t* __restrict__ zpage,
+    float* __restrict__ slab,        // [SPLIT][64][256]
+    int N, int H, int Wp, int K, int Ho, int Wo, int stride, int pad,
+    int split, long m_per_split,
+    unsigned long long magic_wo, unsigned shift_wo,
+    unsigned long long magic_ho, unsigned shift_ho) {
+  constexpr int BK = 64, BC = 256, BM_ = 64;
+  constexpr int WK = BK / 2, WC = BC / 2;
+  constexpr int KF = WK / 16, CF = WC / 16;
+  constexpr int PA = BK / 16 * 2;
+  constexpr int PB = BC / 16 * 2;
+  constexpr int TILE = BM_ * BK;
+  constexpr int BUF = BM_ * (BK + BC);
+  __shared__ __attribute__((aligned(16))) bf16_t smem[2 * BUF];
+
+  const long M = (long)N * Ho * Wo;
+  const int sp = blockIdx.z;
+  const long m0 = (long)sp * m_per_split;
+  const long m1 = (m0 + m_per_split < M) ? (m0 + m_per_split) : M;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc2 = wave & 1;
+
+  const int s_pos = lane >> 3;
+  const int s_mr = (lane >> 1) & 3;
+  const int s_half = lane & 1;
+  const int s_mb = (s_pos < 4) ? (2 * s_pos) : (2 * (s_pos - 4) + 1);
+  const int s_mlocal = s_mb * 4 + s_mr;
+  const int s_koff = s_half * 8;
+
+  const bf16_t* dyrow[2];
+  const bf16_t* xbase[2];
+  int xhb[2];
+  bool dv[2];
+
+  auto decompose = [&](long mbase) {
+    #pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      long m = mbase + g * 32 + s_mlocal;
+      dv[g] = m < m1;
+      dyrow[g] = dv[g] ? (dy + m * K) : zpage;
+      if (dv[g]) {
+        unsigned mu = (unsigned)m;
+        unsigned q1 = st_mdiv(mu, magic_wo, shift_wo);
+        int wo = (int)(mu - q1 * (unsigned)Wo);
+        unsigned n_u = st_mdiv(q1, magic_ho, shift_ho);
+        int ho = (int)(q1 - n_u * (unsigned)Ho);
+        xhb[g] = ho * stride - pad;
+        xbase[g] = x4 + (((long)(int)n_u * H + xhb[g]) * Wp +
+                         (wo * stride - pad) + STEM_HALO) * 4;
+      } else {
+        xhb[g] = -100000;
+        xbase[g] = zpage;
+      }
+    }
+  };
+
+  auto stage = [&](int buf) {
+    bf16_t* ldy = smem + buf * BUF;
+    bf16_t* lx = ldy + TILE;
+    #pragma unroll
+    for (int p = wave; p < PA; p += 4) {
+      const int k16 = p >> 1, g = p & 1;
+      int kk = k16 * 16 + s_koff;  // K == 64, always in range
+      const bf16_t* src = dv[g] ? (dyrow[g] + kk) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src, (LDS_AS void*)(ldy + p * 512), 16, 0, 0);
+    }
+    #pragma unroll
+    for (int p = wave; p < PB; p += 4) {
+      const int c16 = p >> 1, g = p & 1;
+      // flat window element range [c16*16 + s_koff, +8) = 2 c4 pixels:
+      // r = flat/32, s_start = ((flat%32)/4)
+      const int flat = c16 * 16 + s_koff;
+      const int r = flat >> 5;
+      const int s_start = (flat >> 2) & 7;
+      int h = xhb[g] + r;
+      bool ok = dv[g] && h >= 0 && h < H;
+      const bf16_t* src =
+          ok ? (xbase[g] + ((long)r * Wp + s_start) * 4) : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const GLOBAL_AS void*)src, (LDS_AS void*)(lx + p * 512), 16, 0, 0);
+    }
+  };
+
+  s_f32x4_v acc[KF][CF];
+  #pragma unroll
+  for (int a = 0; a < KF; ++a)
+    #pragma unroll
+    for (int b = 0; b < CF; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  const long nsteps = (m1 - m0 + BM_ - 1) / BM_;
+  decompose(m0);
+  stage(0);
+  if (1 < nsteps) decompose(m0 + BM_);
+  __syncthreads();
+
+  int cur = 0;
+  for (long t = 0; t < nsteps; ++t) {
+    if (t + 1 < nsteps) {
+      stage(cur ^ 1);
+      if (t + 2 < nsteps) decompose(m0 + (t + 2) * BM_);
+    }
+    bf16_t* ldy = smem + cur * BUF;
+    bf16_t* lx = ldy + TILE;
+    #pragma unroll
+    for (int mh = 0; mh < 2; ++mh) {
+      s_tr64_t fk0[KF], fk1[KF], fc0[CF], fc1[CF];
+      #pragma unroll
+      for (int a = 0; a < KF; ++a) {
+        const int k16 = (wr * WK) / 16 + a;
+        unsigned addr =
+            (unsigned)(unsigned long long)(const LDS_AS bf16_t*)(
+                ldy + ((k16 * 2 + mh) * 8) * 64) + lane * 8;
+        asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(fk0[a]) : "v"(addr));
+        asm volatile("ds_read_b64_tr_b16 %0, %1 offset:512"
+                     : "=v"(fk1[a]) : "v"(addr));
+      }
+      #pragma unroll
+      for (int b = 0; b < CF; ++b) {
+        const int c16 = (wc2 * WC) / 16 + b;
+        unsigned addr =
+            (unsigned)(unsigned long long)(const LDS_AS bf16_t*)(
+                lx + ((c16 * 2 + mh) * 8) * 64) + lane * 8;
+        asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(fc0[b]) : "v"(addr));
+        asm volatile("ds_read_b64_tr_b16 %0, %1 offset:512"
+                     : "=v"(fc1[b]) : "v"(addr));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      #pragma unroll
+      for (int a = 0; a < KF; ++a)
+        #pragma unroll
+        for (int b = 0; b < CF; ++b) {
+          union { struct { s_tr64_t lo, hi; } p; s_bf16x8_v v; } fa, fb;
+          fa.p.lo = fk0[a]; fa.p.hi = fk1[a];
+          fb.p.lo = fc0[b]; fb.p.hi = fc1[b];
+          acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa.v, fb.v, acc[a][b], 0, 0, 0);
+        }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  float* out = slab + (long)sp * BK * BC;
+  const int d_c = lane & 15;
+  const int d_k0 = (lane >> 4) * 4;
+  #pragma unroll
+  for (int a = 0; a < KF; ++a)
+    #pragma unroll
+    for (int b = 0; b < CF; ++b) {
+      int c = wc2 * WC + b * 16 + d_c;
+      #pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        int k = wr * WK + a * 16 + d_k0 + q;
+        out[(long)k * BC + c] = acc[a][b][q];
+      }
+    }
+}
+
+DDLW_EXPORT int ddlw_stem_wgrad(const void* dy, const void* x4,
+                                const void* zpage, void* slab, void* dw4,
+                                int N, int H, int W, int K, int Ho, int Wo,
+                                int split, void* stream) {
+  if (K != 64) {
+    ddlw_set_error("stem_wgrad: K must be 64");
+    return 2;
+  }
+  const int Wp = W + 2 * STEM_HALO;
+  long M = (long)N * Ho * Wo;
+  unsigned long long mg_wo, mg_ho;
+  unsigned sh_wo, sh_ho;
+  stem_magic((unsigned)Wo, &mg_wo, &sh_wo);
+  stem_magic((unsigned)Ho, &mg_ho, &sh_ho);
+  long m_per_split = (M + split - 1) / split;
+  hipStream_t st = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_stem_wgrad, dim3(1, 1, split), dim3(256), 0, st,
+                     (const bf16_t*)dy, (const bf16_t*)x4,
+                     (const bf16_t*)zpage, (float*)slab, N, H, Wp, K, Ho, Wo,
+                     2, 3, split, m_per_split, mg_wo, sh_wo, mg_ho, sh_ho);
+  {
+    hipError_t err_ = hipGetLastError();
+    if (err_ != hipSuccess) { ddlw_set_error(hipGetErrorString(err_)); return 1; }
+  }
+  long elems = (long)K * 256;
+  long g = (elems + 31) / 32;
+  hipLaunchKernelGGL(k_wgrad_reduce, dim3((int)g), dim3(256), 0, st,
+                     (const float*)slab, (bf16_t*)dw4, elems, split);
+  DDLW_CHECK_LAUNCH();
+}

@@ -401,3 +401,17 @@ def test_eval_bn_fold_parity():
     s = y_ref.float().abs().max() + 1e-6
     rel = ((y_fold.float() - y_ref.float()).abs().max() / s).item()
     assert rel < 5e-2, rel
+
+
+def test_stem_wgrad_parity():
+    """Stem weight gradient (split-K over the c4+halo image) vs fp32 stock."""
+    from ddlw_amd.ops import conv_gemm
+
+    torch.manual_seed(13)
+    x = _cl(torch.randn(5, 3, 64, 64, device=_cuda()).to(torch.bfloat16))
+    dy = _cl(torch.randn(5, 64, 32, 32, device=_cuda()).to(torch.bfloat16))
+    dw = conv_gemm.stem_wgrad_kernel(dy, x).float()
+    ref = torch.nn.grad.conv2d_weight(x.float(), (64, 3, 7, 7), dy.float(),
+                                      stride=2, padding=3)
+    scale = ref.abs().max() + 1e-6
+    assert ((dw - ref).abs().max() / scale).item() < 5e-2
