@@ -297,19 +297,14 @@ DDLW_EXPORT int ddlw_conv_wgrad(const void* dy, const void* x, const void* zpage
                        Ho, Wo, R, S, stride, pad, split, m_per_split, mg_wo,  \
                        sh_wo, mg_ho, sh_ho);                                  \
   } while (0)
-  int wg_bm = 64;
-  {
-    const char* e = getenv("DDLW_WGRAD_BM");
-    if (e && e[0] == '1') wg_bm = 128;       // "128"
-    // default stays 64: the deeper 128-m k-step measured EQUAL OR SLOWER
-    // on every 64x64-tile route (110-234 vs 120-248 TF) — the small-tile
-    // wgrad is staging-bandwidth-bound (AI = 32 flops/B), not barrier- or
-    // pipeline-depth-bound. Documented negative result; env keeps the A/B.
-  }
+  // WG_BM=128 (a deeper 128-m k-step, WLAUNCH3(64,64,128)) measured EQUAL
+  // OR SLOWER on every 64x64-tile route (110-234 vs 120-248 TF): the
+  // small-tile wgrad is staging-bandwidth-bound (AI = 32 flops/B), not
+  // barrier- or pipeline-depth-bound — documented negative result; the
+  // instantiation is dropped (it also spilled 3 SGPRs).
   if (K >= 128 && C >= 128) WLAUNCH(128, 128);
   else if (C >= 128) WLAUNCH(64, 128);
   else if (K >= 128) WLAUNCH(128, 64);
-  else if (wg_bm == 128) WLAUNCH3(64, 64, 128);
   else WLAUNCH(64, 64);
 #undef WLAUNCH
 #undef WLAUNCH3
