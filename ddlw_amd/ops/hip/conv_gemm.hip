@@ -1015,6 +1015,12 @@ static TilePick pick_tile(long M, int K, int C, int T, bool want_stats,
   }
   p.bm = 128;
   p.wvm = 2;
+  // (A BN=256 single-buffer tile for the T==1 K>=256 1x1 layers — halving
+  // B re-reads and block count — measured 111 vs 195 TF and -2.8% on the
+  // whole step: 128 acc VGPRs + 132 AGPRs force 1 block/CU and the
+  // latency-bound single-K-step kernel lives on block-level overlap.
+  // Removed; the 128x128 BUFS=1 tile at 4 blocks/CU is ~73% of its
+  // memory roofline already.)
   p.bn = (K >= 128) ? 128 : (K >= 64 ? 64 : 32);
   // bnb (fused BN-backward reduce) must use the LDS-bounce epilogue on
   // EVERY shape: the direct epilogue's per-element stride-K x loads are
