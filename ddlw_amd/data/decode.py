@@ -90,27 +90,43 @@ class ParallelDecoder:
         pending = {}  # seq -> (start, end)
         next_submit = 0
         done = 0
-        while done < len(chunks):
-            while free and next_submit < len(chunks):
-                s, e = chunks[next_submit]
-                slot = free.pop()
-                pool.task_q.put((next_submit, slot, list(contents[s:e])))
-                pending[next_submit] = (s, e)
-                next_submit += 1
-            try:
-                status, seq, slot, payload = pool.res_q.get(timeout=10.0)
-            except queue.Empty:
-                dead = pool.any_dead()
-                if dead:
-                    raise RuntimeError(f"decode worker(s) died (pids {dead})")
-                continue
-            if status == "err":
-                raise RuntimeError(f"decode failed in worker: {payload}")
-            s, e = pending.pop(seq)
-            out[s:e] = pool.view[slot, : e - s]
-            free.append(slot)
-            done += 1
+        try:
+            while done < len(chunks):
+                while free and next_submit < len(chunks):
+                    s, e = chunks[next_submit]
+                    slot = free.pop()
+                    pool.task_q.put((next_submit, slot, list(contents[s:e])))
+                    pending[next_submit] = (s, e)
+                    next_submit += 1
+                try:
+                    status, seq, slot, payload = pool.res_q.get(timeout=10.0)
+                except queue.Empty:
+                    dead = pool.any_dead()
+                    if dead:
+                        raise RuntimeError(f"decode worker(s) died (pids {dead})")
+                    continue
+                if status == "err":
+                    raise RuntimeError(f"decode failed in worker: {payload}")
+                s, e = pending.pop(seq)
+                out[s:e] = pool.view[slot, : e - s]
+                free.append(slot)
+                done += 1
+        except BaseException:
+            self._drain(pool, next_submit - done)
+            raise
         return torch.from_numpy(out)
+
+    def _drain(self, pool: _ProcDecodePool, outstanding: int) -> None:
+        """Consume in-flight results after an abnormal exit so stale seq
+        ids cannot be attributed to the NEXT map/imap call on this pool."""
+        for _ in range(max(0, outstanding)):
+            try:
+                pool.res_q.get(timeout=10.0)
+            except queue.Empty:
+                # workers wedged or dead — the pool is unusable; rebuild it
+                pool.close()
+                self._pool = None
+                return
 
     def imap(self, contents: Sequence):
         """Pipelined chunk iterator: yields decoded ``chunk_size`` batches
@@ -142,6 +158,7 @@ class ParallelDecoder:
         ready = {}  # seq -> slot
         next_submit = 0
         next_yield = 0
+        received = 0
         prev_slot = None
         try:
             while next_yield < len(chunks):
@@ -158,6 +175,7 @@ class ParallelDecoder:
                         if dead:
                             raise RuntimeError(f"decode worker(s) died (pids {dead})")
                         continue
+                    received += 1
                     if status == "err":
                         raise RuntimeError(f"decode failed in worker: {payload}")
                     ready[seq] = (slot, payload)
@@ -169,8 +187,9 @@ class ParallelDecoder:
                 yield view
                 next_yield += 1
         finally:
-            if prev_slot is not None:
-                free.append(prev_slot)
+            # an abandoned/errored iteration leaves results in flight; drain
+            # them so the pool can serve the next call with fresh seq ids
+            self._drain(pool, next_submit - received)
 
     def close(self) -> None:
         if self._pool is not None:

@@ -204,3 +204,18 @@ def test_parallel_decoder_imap_order_and_content():
     with ParallelDecoder(_times_two, workers=2, chunk_size=3) as dec:
         sizes = [c.shape[0] for c in dec.imap(rows[:7])]
     assert sizes == [3, 3, 1]
+
+
+def test_parallel_decoder_survives_abandoned_imap():
+    """Breaking out of imap mid-stream must not corrupt the next call
+    (stale in-flight results are drained)."""
+    from ddlw_amd.data.decode import ParallelDecoder
+
+    rows = [bytes([i, i + 1, i + 2, i + 3]) for i in range(0, 80, 4)]
+    serial = ParallelDecoder(_times_two, workers=0).map(rows)
+    with ParallelDecoder(_times_two, workers=3, chunk_size=2) as dec:
+        for i, chunk in enumerate(dec.imap(rows)):
+            if i == 2:
+                break  # abandon mid-stream with tasks in flight
+        out = dec.map(rows)  # pool reuse must still be correct
+    assert torch.equal(out, serial)
