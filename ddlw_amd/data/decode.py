@@ -89,6 +89,7 @@ class ParallelDecoder:
         free = list(range(pool.slots))
         pending = {}  # seq -> (start, end)
         next_submit = 0
+        received = 0
         done = 0
         try:
             while done < len(chunks):
@@ -105,6 +106,7 @@ class ParallelDecoder:
                     if dead:
                         raise RuntimeError(f"decode worker(s) died (pids {dead})")
                     continue
+                received += 1
                 if status == "err":
                     raise RuntimeError(f"decode failed in worker: {payload}")
                 s, e = pending.pop(seq)
@@ -112,7 +114,7 @@ class ParallelDecoder:
                 free.append(slot)
                 done += 1
         except BaseException:
-            self._drain(pool, next_submit - done)
+            self._drain(pool, next_submit - received)
             raise
         return torch.from_numpy(out)
 
