@@ -322,7 +322,7 @@ class ShardedParquetLoader:
         finally:
             pool.shutdown(wait=False)
 
-    def _batches_proc(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor, Callable]]:
+    def _batches_proc(self, owner: Optional[dict] = None) -> Iterator[Tuple[torch.Tensor, torch.Tensor, Callable]]:
         """Decode batches via the process pool. Yields
         ``(images_view, labels, release)`` where ``images_view`` is a
         zero-copy torch view over a shared-memory slot: the consumer must
@@ -343,6 +343,11 @@ class ShardedParquetLoader:
             self.transform, sample.shape, sample.dtype, self.batch_size,
             self.workers,
             pin=self.device is not None and self.device.type == "cuda")
+        if owner is not None:
+            # the consumer owns teardown: the pinned shm ring must outlive
+            # its in-flight H2D DMA, which only the consumer can order
+            # (generator exhaustion happens BEFORE the last stage is issued)
+            owner["pool"] = pool
         self._pool_info = (pool.slots, pool.pinned)
         free_slots: deque = deque(range(pool.slots))
         pending_labels = {}
@@ -385,7 +390,8 @@ class ShardedParquetLoader:
                 yield imgs, labels, (lambda s=slot: free_slots.append(s))
                 yield_seq += 1
         finally:
-            pool.close()
+            if owner is None:
+                pool.close()
 
     def _resolve_pool(self) -> str:
         if self.pool != "auto":
@@ -427,10 +433,12 @@ class ShardedParquetLoader:
                     continue
             return False
 
+        pool_owner: dict = {}
+
         def producer():
             try:
                 if mode == "process":
-                    for imgs, labels, release in self._batches_proc():
+                    for imgs, labels, release in self._batches_proc(pool_owner):
                         if not _bounded_put((imgs, labels, release)):
                             release()
                             return
@@ -484,13 +492,14 @@ class ShardedParquetLoader:
                 dl.record_stream(cs)
                 yield di, dl
         finally:
-            stop.set()
-            # in-flight DMA reads the shm ring — finish it before the
-            # producer (and with it the pool / hostUnregister) tears down
+            # order matters: finish in-flight DMA from the pinned shm ring
+            # BEFORE the pool (hostUnregister + unmap) can tear down — the
+            # consumer owns the pool exactly for this reason
             if pending:
                 torch.cuda.synchronize(self.device)
                 while pending:
                     pending.popleft()[1]()
+            stop.set()
             # drain so the producer can exit, then join it so no thread is
             # left inside native code at interpreter teardown
             while not q.empty():
@@ -501,6 +510,11 @@ class ShardedParquetLoader:
                 except queue.Empty:
                     break
             t.join(timeout=10.0)
+            pool = pool_owner.get("pool")
+            if pool is not None:
+                if pool.pinned:
+                    torch.cuda.synchronize(self.device)
+                pool.close()
 
 
 class Converter:
