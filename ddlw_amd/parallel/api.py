@@ -230,11 +230,11 @@ class DistributedOptimizer:
         bucket_cap_mb: float = 32.0,
         average: bool = True,
     ):
-        from ..utils.trace import ChromeTracer
+        from ..utils.trace import get_tracer
 
         self.optimizer = optimizer
         self.average = average
-        self._tracer = ChromeTracer()  # DDLW_TIMELINE collective events
+        self._tracer = get_tracer()  # DDLW_TIMELINE collective events
         self._params: List[torch.nn.Parameter] = [
             p for g in optimizer.param_groups for p in g["params"] if p.requires_grad
         ]

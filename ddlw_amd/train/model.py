@@ -23,7 +23,7 @@ import torch.nn.functional as F
 
 from ..core import tracking
 from ..core.model_io import log_model
-from ..utils.trace import ChromeTracer
+from ..utils.trace import ChromeTracer, get_tracer
 from .callbacks import Callback, MetricAverageCallback
 
 _autolog_enabled = False
@@ -209,7 +209,7 @@ class Model:
         # Horovod-Timeline equivalent: DDLW_TIMELINE=<path> -> chrome trace
         from ..parallel import api as _api
 
-        tracer = ChromeTracer()
+        tracer = get_tracer()
         data_iter = iter(data)
         for epoch in range(epochs):
             for cb in callbacks:

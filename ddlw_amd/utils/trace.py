@@ -65,3 +65,16 @@ class ChromeTracer:
         with self._lock:
             with open(self.path, "w") as f:
                 json.dump({"traceEvents": self._events}, f)
+
+
+_shared: Optional[ChromeTracer] = None
+
+
+def get_tracer() -> ChromeTracer:
+    """Process-wide shared tracer: every subsystem (train loop spans,
+    DistributedOptimizer collective events) appends to ONE timeline file —
+    separate instances on the same path would overwrite each other."""
+    global _shared
+    if _shared is None:
+        _shared = ChromeTracer()
+    return _shared
