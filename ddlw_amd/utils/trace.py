@@ -73,8 +73,11 @@ _shared: Optional[ChromeTracer] = None
 def get_tracer() -> ChromeTracer:
     """Process-wide shared tracer: every subsystem (train loop spans,
     DistributedOptimizer collective events) appends to ONE timeline file —
-    separate instances on the same path would overwrite each other."""
+    separate instances on the same path would overwrite each other. Rebuilt
+    when DDLW_TIMELINE changes (tests toggle it per-case)."""
     global _shared
-    if _shared is None:
+    env = os.environ.get("DDLW_TIMELINE")
+    if _shared is None or _shared._env_path != env:
         _shared = ChromeTracer()
+        _shared._env_path = env
     return _shared
