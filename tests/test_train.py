@@ -39,12 +39,14 @@ def test_evaluate_returns_loss_and_metrics(ddlw_home):
 
 def test_early_stopping_stops(ddlw_home):
     m = Model(build_small_cnn(16, 16, num_classes=3)).compile("SGD", learning_rate=0.0)
-    # lr=0 -> no improvement -> stops after patience+1 epochs without progress
+    # lr=0 -> no real improvement. min_delta must dominate the tiny val_loss
+    # drift from BatchNorm running-stat updates (which continue at lr=0 and
+    # can micro-"improve" for many epochs — was a rare flake at 1e-9).
     hist = m.fit(
         _toy_data(),
         epochs=10,
         validation_data=_toy_data(seed=1),
-        callbacks=[EarlyStopping(monitor="val_loss", min_delta=1e-9, patience=1)],
+        callbacks=[EarlyStopping(monitor="val_loss", min_delta=0.5, patience=1)],
         verbose=0,
     )
     assert len(hist.history["loss"]) < 10
