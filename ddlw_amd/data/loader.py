@@ -42,6 +42,44 @@ from .preprocess import preprocess_bytes
 from ..core.config import current_setup
 
 
+def _default_transform(content, img_height: int, img_width: int):
+    # module-level (picklable) so the decode pool can use forkserver
+    return preprocess_bytes(content, img_height, img_width)
+
+
+_forkserver_preloaded = False
+
+
+def _pool_ctx(transform):
+    """Start-method choice for the decode pool. Plain fork of a
+    CUDA-initialized parent can wedge the child (forked while another
+    thread holds a runtime/allocator lock) — children then sit ALIVE but
+    dead, observed as a silent hang late in big GPU test sessions. When
+    CUDA is up and the transform is picklable, use a forkserver (workers
+    fork from a small clean server; the server preloads this module once
+    so each worker skips the torch import)."""
+    import multiprocessing as mp
+    import pickle
+
+    try:
+        cuda_up = torch.cuda.is_initialized()
+    except Exception:
+        cuda_up = False
+    if cuda_up and "forkserver" in mp.get_all_start_methods():
+        try:
+            pickle.dumps(transform)
+            ctx = mp.get_context("forkserver")
+            global _forkserver_preloaded
+            if not _forkserver_preloaded:
+                ctx.set_forkserver_preload(["ddlw_amd.data.loader"])
+                _forkserver_preloaded = True
+            return ctx
+        except Exception:
+            pass
+    method = "fork" if "fork" in mp.get_all_start_methods() else "spawn"
+    return mp.get_context(method)
+
+
 def shard_row_groups(num_row_groups: int, cur_shard: int, shard_count: int) -> List[int]:
     """Row groups owned by shard ``cur_shard`` of ``shard_count``:
     disjoint across shards, exhaustive, round-robin (rank r gets r, r+W, ...)."""
@@ -105,11 +143,9 @@ class _ProcDecodePool:
 
     def __init__(self, transform, sample_shape, sample_dtype, batch_size: int,
                  workers: int, pin: bool = False):
-        import multiprocessing as mp
         from multiprocessing import shared_memory
 
-        method = "fork" if "fork" in mp.get_all_start_methods() else "spawn"
-        ctx = mp.get_context(method)
+        ctx = _pool_ctx(transform)
         self.workers = max(1, int(workers))
         self.slot_shape = (int(batch_size),) + tuple(int(s) for s in sample_shape)
         self.np_dtype = np.dtype(sample_dtype)
@@ -250,9 +286,11 @@ class ShardedParquetLoader:
         self.device = device
         self.content_column = content_column
         self.label_column = label_column
-        self.transform = transform or (
-            lambda c: preprocess_bytes(c, self.img_height, self.img_width)
-        )
+        import functools
+
+        self.transform = transform or functools.partial(
+            _default_transform, img_height=self.img_height,
+            img_width=self.img_width)
         self.prefetch = prefetch
         if pool not in ("auto", "process", "thread"):
             raise ValueError(f"pool must be auto|process|thread, got {pool!r}")
@@ -371,6 +409,7 @@ class ShardedParquetLoader:
                         exhausted = True
                 if exhausted and yield_seq == submit_seq:
                     return
+                stall = 0.0
                 while not heap or heap[0][0] != yield_seq:
                     try:
                         status, seq, slot, payload = pool.res_q.get(timeout=5.0)
@@ -379,7 +418,15 @@ class ShardedParquetLoader:
                         if dead:
                             raise RuntimeError(
                                 f"decode worker(s) died (pids {dead})")
+                        stall += 5.0
+                        if stall >= 120.0:
+                            # alive-but-wedged workers (fork-after-CUDA
+                            # hazard): fail loudly instead of hanging
+                            raise RuntimeError(
+                                "decode pool made no progress for 120s "
+                                "(workers alive but wedged)")
                         continue
+                    stall = 0.0
                     if status == "err":
                         raise RuntimeError(f"decode failed in worker: {payload}")
                     heapq.heappush(heap, (seq, slot, payload))
