@@ -494,10 +494,12 @@ class ShardedParquetLoader:
                         if not _bounded_put((batch[0], batch[1], None)):
                             return
             finally:
-                try:
-                    q.put_nowait(None)
-                except queue.Full:
-                    pass
+                # the end-of-stream sentinel must ARRIVE: a put_nowait here
+                # is silently dropped when the bounded queue is full (fast
+                # decode pool, consumer still staging) and the consumer
+                # then blocks on q.get() forever — the late-session GPU
+                # hang. Bounded-put it like any other item.
+                _bounded_put(None)
 
         t = threading.Thread(target=producer, daemon=True)
         t.start()
