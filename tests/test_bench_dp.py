@@ -30,15 +30,16 @@ def _free_port() -> int:
     return port
 
 
-def _run_bench_dp(tmp_path, extra_args=()):
+def _run_bench_dp(tmp_path, extra_args=(), nproc=2):
     env = dict(os.environ)
     env["DDLW_HOME"] = str(tmp_path)
     env["MASTER_ADDR"] = "127.0.0.1"
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", "--nproc-per-node", "2",
+        "--nnodes=1", "--nproc-per-node", str(nproc),
         "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
-        str(REPO / "bench.py"), "--gpus", "2", "--steps", "2", "--warmup", "1",
+        str(REPO / "bench.py"), "--gpus", str(nproc),
+        "--steps", "2", "--warmup", "1",
         *extra_args,
     ]
     res = subprocess.run(cmd, capture_output=True, text=True, timeout=900,
@@ -55,6 +56,15 @@ def test_bench_dp_ws2_exact_code_path(tmp_path):
     assert out["config"]["parallelism"] == "dp2"
     assert out["config"]["global_batch"] == out["config"]["per_gpu_batch"] * 2
     assert out["value"] > 0
+
+
+def test_bench_dp_ws4_exact_code_path(tmp_path):
+    """World 4 on gloo: >2 ranks exercise uneven bucket splits and the
+    coalesced broadcast at the driver's larger-N geometry."""
+    out = _run_bench_dp(tmp_path, nproc=4)
+    assert out["n_gpus"] == 4
+    assert out["config"]["parallelism"] == "dp4"
+    assert out["config"]["global_batch"] == out["config"]["per_gpu_batch"] * 4
 
 
 def _overlap_worker(seed):
