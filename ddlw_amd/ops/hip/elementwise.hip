@@ -428,19 +428,24 @@ __global__ __launch_bounds__(256) void k_maxpool3x3s2_fwd(
 
 // backward: gather — for each input element, sum dy over the <=4 output
 // windows that could have selected it, checking the recorded argmax.
+// grid: y = (n*H + h) input row, x covers w * vecC — one vector per
+// thread, addresses by shift/mask (vecC is a power of two for every
+// ResNet/MobileNet C) instead of the per-iteration div/mod chains of the
+// old grid-stride form (which dominated this latency-bound kernel)
 __global__ __launch_bounds__(256) void k_maxpool3x3s2_bwd(
     const bf16_t* __restrict__ dy, const unsigned char* __restrict__ argmax,
     bf16_t* __restrict__ dx,
-    int N, int H, int W, int C, int Ho, int Wo) {
+    int N, int H, int W, int C, int Ho, int Wo, int lv /* log2(C/8) */) {
   const int vecC = C >> 3;
-  const unsigned total = (unsigned)((long)N * H * W * vecC);
-  for (unsigned i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += gridDim.x * blockDim.x) {
-    unsigned vc = i % (unsigned)vecC;
-    unsigned t = i / (unsigned)vecC;
-    int w = (int)(t % (unsigned)W); t /= (unsigned)W;
-    int h = (int)(t % (unsigned)H); t /= (unsigned)H;
-    int n = (int)t;
+  {
+    const unsigned idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (unsigned)(W << lv)) return;
+    const int w = (int)(idx >> lv);
+    const unsigned vc = idx & ((1u << lv) - 1);
+    const int h = blockIdx.y % (unsigned)H;
+    const int n = blockIdx.y / (unsigned)H;
+    const unsigned long long i =
+        (((unsigned long long)blockIdx.y * W + w) << lv) + vc;
     float acc[8] = {0};
     // output windows covering (h, w): ho*2-1 <= h <= ho*2+1
     int ho_lo = (h - 1 + 1) / 2, ho_hi = (h + 1) / 2;  // ceil((h-1)/2), floor((h+1)/2)
@@ -472,6 +477,7 @@ __global__ __launch_bounds__(256) void k_maxpool3x3s2_bwd(
     *reinterpret_cast<uint4*>(dx + (long)i * 8) = o.v;
   }
 }
+
 
 // ---------------------------------------------------------------------------
 // Global average pooling NHWC: y[n,c] = mean_hw x[n,h,w,c]  (K6).
@@ -900,14 +906,22 @@ DDLW_EXPORT int ddlw_maxpool3x3s2_fwd(const void* x, void* y, void* argmax,
 DDLW_EXPORT int ddlw_maxpool3x3s2_bwd(const void* dy, const void* argmax, void* dx,
                                       int N, int H, int W, int C, int Ho, int Wo,
                                       void* stream) {
-  long total = (long)N * H * W * (C >> 3);
-  if (total >= (1ll << 31)) {
-    ddlw_set_error("maxpool3x3s2: tensor too large for 32-bit indexing");
+  int vecC = C >> 3;
+  if (vecC <= 0 || (vecC & (vecC - 1)) != 0) {
+    ddlw_set_error("maxpool3x3s2_bwd: C/8 must be a power of two");
     return 2;
   }
-  hipLaunchKernelGGL(k_maxpool3x3s2_bwd, dim3(grid_1d(total)), dim3(256), 0,
+  if ((long)N * H >= 65536 || (long)H * W * vecC >= (1ll << 31)) {
+    ddlw_set_error("maxpool3x3s2: tensor too large for the 2D grid");
+    return 2;
+  }
+  int lv = 0;
+  while ((1 << lv) < vecC) ++lv;
+  dim3 grid(((W << lv) + 255) / 256, N * H);
+  hipLaunchKernelGGL(k_maxpool3x3s2_bwd, grid, dim3(256), 0,
                      (hipStream_t)stream, (const bf16_t*)dy,
-                     (const unsigned char*)argmax, (bf16_t*)dx, N, H, W, C, Ho, Wo);
+                     (const unsigned char*)argmax, (bf16_t*)dx, N, H, W, C, Ho,
+                     Wo, lv);
   DDLW_CHECK_LAUNCH();
 }
 
