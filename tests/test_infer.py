@@ -219,3 +219,22 @@ def test_parallel_decoder_survives_abandoned_imap():
                 break  # abandon mid-stream with tasks in flight
         out = dec.map(rows)  # pool reuse must still be correct
     assert torch.equal(out, serial)
+
+
+def test_predict_table_from_parquet_dir(ddlw_home, tmp_path):
+    """predict_table accepts a Parquet dataset DIRECTORY (bronze/silver
+    layout) as the input side of the withColumn contract."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from ddlw_amd.infer import predict_table
+
+    uri = _package(ddlw_home)
+    contents, labels = make_synthetic_dataset(8, 16, 16, num_classes=5, seed=8, jpeg=True)
+    d = tmp_path / "ds"
+    d.mkdir()
+    t = pa.table({"content": pa.array(contents, pa.binary()), "label_idx": labels})
+    pq.write_table(t.slice(0, 4), d / "part-0.parquet")
+    pq.write_table(t.slice(4, 4), d / "part-1.parquet")
+    out = predict_table(uri, str(d), num_workers=2, gpus=[])
+    assert out.num_rows == 8 and "prediction" in out.column_names

@@ -175,3 +175,19 @@ def _boom(content):
     if mp.current_process().daemon:
         raise ValueError("bad jpeg")
     return np.zeros((4, 4, 3), np.uint8)
+
+
+def test_loader_finite_multi_epoch_row_count(ddlw_home):
+    """num_epochs=2 yields exactly 2x the rows, thread and process pools."""
+    tbl = _make_table(24, seed=6)
+    conv = make_converter(tbl, row_group_rows=8)
+    for pool in ("thread", "process"):
+        n = 0
+        with conv.make_torch_dataset(
+            batch_size=5, num_epochs=2, img_height=16, img_width=16,
+            pool=pool, workers_count=2,
+        ) as loader:
+            for imgs, labels in loader:
+                n += labels.numel()
+        assert n == 48, (pool, n)
+    conv.delete()
