@@ -118,20 +118,26 @@ class Runner:
 
         result: Optional[tuple] = None
         try:
-            # rank 0's result arrives on the queue; join everyone
-            deadline_failed = False
-            for p in procs:
-                p.join(self.timeout_s)
-                if p.is_alive():
-                    deadline_failed = True
+            # poll the gang: ANY worker death aborts promptly (a sequential
+            # join would wait the earlier ranks' full timeout before
+            # noticing a later rank died while rank 0 blocks in a
+            # collective); one shared deadline for the whole gang
+            import time as _time
+
+            deadline = _time.monotonic() + self.timeout_s
+            while True:
+                for p in procs:
+                    if p.exitcode not in (0, None):
+                        raise RuntimeError(
+                            f"worker (pid {p.pid}) exited with code "
+                            f"{p.exitcode}; aborting job"
+                        )
+                if not any(p.is_alive() for p in procs):
                     break
-                if p.exitcode not in (0, None):
-                    # failure detection: abort the gang
-                    raise RuntimeError(
-                        f"worker (pid {p.pid}) exited with code {p.exitcode}; aborting job"
-                    )
-            if deadline_failed:
-                raise TimeoutError(f"workers still alive after {self.timeout_s}s; aborting")
+                if _time.monotonic() > deadline:
+                    raise TimeoutError(
+                        f"workers still alive after {self.timeout_s}s; aborting")
+                _time.sleep(0.2)
             if not result_q.empty():
                 result = result_q.get()
         finally:
