@@ -28,3 +28,20 @@ def test_bench_json_contract(tmp_path):
     assert out["scaling"] == "weak"
     assert out["higher_is_better"] is True
     assert "global_batch" in out["config"] and "parallelism" in out["config"]
+
+
+def test_conv_dispatch_table_schema():
+    """The measured router table must stay well-formed: every entry maps
+    fwd/dgrad/wgrad to 'hip' or 'stock' (a corrupted table would silently
+    route everything to the library)."""
+    import json
+    from pathlib import Path
+
+    table = json.loads(
+        (REPO / "ddlw_amd" / "ops" / "conv_dispatch.json").read_text())
+    assert len(table) >= 20  # the 22 ResNet-50 body shapes
+    for key, ent in table.items():
+        assert set(ent) == {"fwd", "dgrad", "wgrad"}, key
+        assert all(v in ("hip", "stock") for v in ent.values()), key
+    # fwd must be fully hip (measured winner on all 22 routes)
+    assert all(e["fwd"] == "hip" for e in table.values())
